@@ -1,0 +1,129 @@
+"""GPU slot scheduler for one remote 8×MI355X node.
+
+New component with no reference counterpart: the
+reference executes tasks on the remote with whatever GPU visibility the
+login shell has.  Here every task acquires a *slot* before it is
+submitted, and the remote command is launched with
+``HIP_VISIBLE_DEVICES=<gpu>`` (plus ``ROCR_VISIBLE_DEVICES`` for
+runtimes that consult it) so concurrent electrons land on distinct
+MI355X GPUs.  With 288 GB HBM3E per GPU there is no reason to co-locate
+two electrons on one device; a slot is exactly one GPU by default
+(``slots_per_gpu`` can oversubscribe for small tasks).
+
+Accounting is module-level and keyed per endpoint so that multiple
+SSHExecutor instances pointing at the same host share one slot table
+(SURVEY.md §7 "Hard parts").  Acquisition is FIFO-fair (asyncio.Condition)
+and prefers the least-recently-released GPU (round-robin) so work spreads
+across the node even at low concurrency.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import itertools
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+
+@dataclass
+class Slot:
+    """A held GPU slot.  ``gpu_id`` indexes the node's physical GPUs."""
+
+    gpu_id: int
+    table: "SlotTable"
+    _released: bool = False
+
+    def env(self) -> Dict[str, str]:
+        return {
+            "HIP_VISIBLE_DEVICES": str(self.gpu_id),
+            "ROCR_VISIBLE_DEVICES": str(self.gpu_id),
+        }
+
+    async def release(self) -> None:
+        if not self._released:
+            self._released = True
+            await self.table._release(self.gpu_id)
+
+    async def __aenter__(self) -> "Slot":
+        return self
+
+    async def __aexit__(self, *exc) -> None:
+        await self.release()
+
+
+class SlotTable:
+    """Per-endpoint table of GPU slots."""
+
+    def __init__(self, num_gpus: int = 8, slots_per_gpu: int = 1):
+        if num_gpus < 1:
+            raise ValueError("num_gpus must be >= 1")
+        self.num_gpus = num_gpus
+        self.slots_per_gpu = slots_per_gpu
+        # free_count[gpu] = remaining capacity; order = round-robin queue of
+        # gpu ids, least-recently used first.
+        self._free: List[int] = [slots_per_gpu] * num_gpus
+        self._order: List[int] = list(range(num_gpus))
+        self._cond: Optional[asyncio.Condition] = None
+        self._in_use = 0
+
+    def _condition(self) -> asyncio.Condition:
+        if self._cond is None:
+            self._cond = asyncio.Condition()
+        return self._cond
+
+    @property
+    def in_use(self) -> int:
+        return self._in_use
+
+    @property
+    def capacity(self) -> int:
+        return self.num_gpus * self.slots_per_gpu
+
+    async def acquire(self, timeout: Optional[float] = None) -> Slot:
+        cond = self._condition()
+
+        async def _take() -> Slot:
+            async with cond:
+                while True:
+                    for idx, gpu in enumerate(self._order):
+                        if self._free[gpu] > 0:
+                            self._free[gpu] -= 1
+                            # Move to the back: next acquire prefers others.
+                            self._order.append(self._order.pop(idx))
+                            self._in_use += 1
+                            return Slot(gpu_id=gpu, table=self)
+                    await cond.wait()
+
+        if timeout is None:
+            return await _take()
+        return await asyncio.wait_for(_take(), timeout=timeout)
+
+    async def _release(self, gpu_id: int) -> None:
+        cond = self._condition()
+        async with cond:
+            self._free[gpu_id] += 1
+            self._in_use -= 1
+            cond.notify_all()
+
+
+# ---------------------------------------------------------------------------
+# Module-level registry, keyed per endpoint (host, user) — shared across
+# executor instances.
+# ---------------------------------------------------------------------------
+
+_tables: Dict[Tuple[str, ...], SlotTable] = {}
+
+
+def get_slot_table(
+    key: Tuple[str, ...], num_gpus: int = 8, slots_per_gpu: int = 1
+) -> SlotTable:
+    table = _tables.get(key)
+    if table is None:
+        table = SlotTable(num_gpus=num_gpus, slots_per_gpu=slots_per_gpu)
+        _tables[key] = table
+    return table
+
+
+def reset() -> None:
+    """Test hook: drop all slot tables."""
+    _tables.clear()

@@ -1,0 +1,306 @@
+// CDNA4 (gfx950 / MI355X) GPU library for the covalent SSH executor.
+//
+// New MI355X-native component with no reference counterpart (the
+// reference plugin, /root/reference/covalent_ssh_plugin, has zero native
+// code — SURVEY.md §2.4).  Provides, behind a plain C ABI consumed via
+// ctypes (so the remote stub needs no torch and no Python extension ABI):
+//
+//   * csp_probe_json  — device probe + measured HBM bandwidth and bf16
+//                       MFMA throughput from the two hand-written
+//                       warm-up kernels below,
+//   * csp_warmup      — clock/cache warm-up (MFMA spin + HBM sweep)
+//                       bounded by a millisecond budget,
+//   * csp_host_alloc / csp_host_free / csp_staging_get /
+//     csp_memcpy_d2h / csp_memcpy_h2d
+//                     — hipHostMalloc-pinned staging for large tensor
+//                       results on the SFTP/stdout return path.
+//
+// Kernel design notes (per /opt/skills/guides/MI355X_MICROARCH.md):
+//   - wave64; 256-thread blocks = 4 waves;
+//   - the MFMA spin uses v_mfma_f32_32x32x16_bf16 (32768 FLOP/instr,
+//     ~32 cyc back-to-back issue per SIMD) with 4 independent
+//     accumulators per wave so issue is never dependency-stalled;
+//   - the HBM sweep reads+writes float4 (16 B/lane) grid-stride with
+//     >=1024 workgroups so all 8 XCDs are saturated;
+//   - gfx950-only: built with --offload-arch=gfx950, no other targets.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <cstdio>
+#include <cstring>
+#include <mutex>
+
+// ---------------------------------------------------------------------------
+// Error plumbing
+// ---------------------------------------------------------------------------
+
+static thread_local char g_err[512];
+
+static int set_err(const char* what, hipError_t e) {
+    snprintf(g_err, sizeof(g_err), "%s: %s", what, hipGetErrorString(e));
+    return -1;
+}
+
+#define HIP_TRY(expr)                                    \
+    do {                                                 \
+        hipError_t _e = (expr);                          \
+        if (_e != hipSuccess) return set_err(#expr, _e); \
+    } while (0)
+
+extern "C" const char* csp_last_error() { return g_err; }
+
+// ---------------------------------------------------------------------------
+// Kernels
+// ---------------------------------------------------------------------------
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+// MFMA spin: each wave issues `iters` v_mfma_f32_32x32x16_bf16 across 4
+// independent accumulators (issue-rate bound, no dependency stall).
+__global__ __launch_bounds__(256) void csp_mfma_spin_kernel(
+    float* __restrict__ out, int iters) {
+    bf16x8 a, b;
+    const int lane = threadIdx.x;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+        a[i] = (__bf16)((lane + i) & 7);
+        b[i] = (__bf16)((lane * 3 + i) & 7);
+    }
+    f32x16 acc0 = {}, acc1 = {}, acc2 = {}, acc3 = {};
+    for (int i = 0; i < iters; i += 4) {
+        acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc0, 0, 0, 0);
+        acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc1, 0, 0, 0);
+        acc2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc2, 0, 0, 0);
+        acc3 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc3, 0, 0, 0);
+    }
+    float s = 0.f;
+#pragma unroll
+    for (int i = 0; i < 16; ++i) s += acc0[i] + acc1[i] + acc2[i] + acc3[i];
+    if (out && s == -1.0f) out[blockIdx.x] = s;  // never true: defeats DCE only
+}
+
+// HBM sweep: grid-stride float4 copy (read 16 B + write 16 B per lane per
+// element).  Doubles as the bandwidth probe and the HBM warm-up.
+__global__ __launch_bounds__(256) void csp_hbm_sweep_kernel(
+    const float4* __restrict__ src, float4* __restrict__ dst, size_t n4) {
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (; i < n4; i += stride) {
+        float4 v = src[i];
+        dst[i] = v;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Device / measurement helpers
+// ---------------------------------------------------------------------------
+
+extern "C" int csp_device_count() {
+    int n = 0;
+    hipError_t e = hipGetDeviceCount(&n);
+    if (e != hipSuccess) {
+        set_err("hipGetDeviceCount", e);
+        return -1;
+    }
+    return n;
+}
+
+static int run_mfma_spin(int iters_per_wave, int blocks, float* ms_out) {
+    hipEvent_t t0, t1;
+    HIP_TRY(hipEventCreate(&t0));
+    HIP_TRY(hipEventCreate(&t1));
+    HIP_TRY(hipEventRecord(t0));
+    hipLaunchKernelGGL(csp_mfma_spin_kernel, dim3(blocks), dim3(256), 0, 0,
+                       nullptr, iters_per_wave);
+    HIP_TRY(hipEventRecord(t1));
+    HIP_TRY(hipEventSynchronize(t1));
+    HIP_TRY(hipEventElapsedTime(ms_out, t0, t1));
+    HIP_TRY(hipEventDestroy(t0));
+    HIP_TRY(hipEventDestroy(t1));
+    return 0;
+}
+
+static int run_hbm_sweep(float4* buf_a, float4* buf_b, size_t n4, int reps,
+                         float* ms_out) {
+    hipEvent_t t0, t1;
+    HIP_TRY(hipEventCreate(&t0));
+    HIP_TRY(hipEventCreate(&t1));
+    HIP_TRY(hipEventRecord(t0));
+    for (int r = 0; r < reps; ++r) {
+        // ping-pong so reads are never served from a just-written line
+        const float4* s = (r & 1) ? buf_b : buf_a;
+        float4* d = (r & 1) ? buf_a : buf_b;
+        hipLaunchKernelGGL(csp_hbm_sweep_kernel, dim3(4096), dim3(256), 0, 0,
+                           s, d, n4);
+    }
+    HIP_TRY(hipEventRecord(t1));
+    HIP_TRY(hipEventSynchronize(t1));
+    HIP_TRY(hipEventElapsedTime(ms_out, t0, t1));
+    HIP_TRY(hipEventDestroy(t0));
+    HIP_TRY(hipEventDestroy(t1));
+    return 0;
+}
+
+// FLOP per v_mfma_f32_32x32x16_bf16: 2*32*32*16
+static const double kFlopPerMfma = 32768.0;
+
+extern "C" int csp_warmup(int device, int budget_ms) {
+    HIP_TRY(hipSetDevice(device));
+    if (budget_ms <= 0) budget_ms = 50;
+    // One short HBM touch (256 MiB) + an MFMA spin sized to the budget.
+    size_t bytes = 256u * 1024u * 1024u;
+    size_t n4 = bytes / sizeof(float4);
+    float4 *a = nullptr, *b = nullptr;
+    HIP_TRY(hipMalloc(&a, bytes));
+    HIP_TRY(hipMalloc(&b, bytes));
+    HIP_TRY(hipMemsetAsync(a, 1, bytes));
+    float ms = 0.f;
+    int rc = run_hbm_sweep(a, b, n4, 2, &ms);
+    if (rc == 0) {
+        // ~40 TFLOP per budget at low clocks -> iters sized for ~budget_ms
+        // at an assumed sub-peak 1 PF/s warm-from-idle rate.
+        double flops_target = 1.0e12 * (budget_ms / 1000.0) * 2.0;
+        int blocks = 1024;
+        double waves = (double)blocks * 4.0;
+        int iters = (int)(flops_target / (waves * kFlopPerMfma));
+        if (iters < 64) iters = 64;
+        rc = run_mfma_spin(iters, blocks, &ms);
+    }
+    (void)hipFree(a);
+    (void)hipFree(b);
+    return rc;
+}
+
+extern "C" int csp_probe_json(int device, char* buf, size_t buflen) {
+    HIP_TRY(hipSetDevice(device));
+    hipDeviceProp_t props;
+    HIP_TRY(hipGetDeviceProperties(&props, device));
+    size_t mem_free = 0, mem_total = 0;
+    HIP_TRY(hipMemGetInfo(&mem_free, &mem_total));
+
+    // --- measured HBM bandwidth: 1 GiB ping-pong copy ------------------
+    size_t bytes = 1024u * 1024u * 1024u;
+    size_t n4 = bytes / sizeof(float4);
+    float4 *a = nullptr, *b = nullptr;
+    HIP_TRY(hipMalloc(&a, bytes));
+    HIP_TRY(hipMalloc(&b, bytes));
+    HIP_TRY(hipMemsetAsync(a, 1, bytes));
+    float warm_ms = 0.f, ms = 0.f;
+    int rc = run_hbm_sweep(a, b, n4, 2, &warm_ms);  // warm
+    if (rc == 0) rc = run_hbm_sweep(a, b, n4, 6, &ms);
+    (void)hipFree(a);
+    (void)hipFree(b);
+    if (rc != 0) return rc;
+    // each rep reads + writes `bytes`
+    double hbm_gbps = (2.0 * (double)bytes * 6.0 / 1.0e9) / ((double)ms / 1000.0);
+
+    // --- measured bf16 MFMA throughput ---------------------------------
+    const int blocks = 1024;  // 4x CU count: fills all 8 XCDs
+    const int iters = 8192;   // per wave
+    rc = run_mfma_spin(iters / 4, blocks, &warm_ms);  // warm clocks
+    if (rc != 0) return rc;
+    rc = run_mfma_spin(iters, blocks, &ms);
+    if (rc != 0) return rc;
+    double waves = (double)blocks * 4.0;
+    double tflops = waves * (double)iters * kFlopPerMfma / ((double)ms / 1000.0) / 1.0e12;
+
+    int written = snprintf(
+        buf, buflen,
+        "{\"name\":\"%s\",\"gcn_arch\":\"%s\",\"device\":%d,"
+        "\"cu_count\":%d,\"max_clock_mhz\":%d,\"lds_per_cu_kb\":%zu,"
+        "\"wavefront_size\":%d,\"hbm_total_gb\":%.1f,\"hbm_free_gb\":%.1f,"
+        "\"hbm_bw_gbps\":%.1f,\"mfma_bf16_tflops\":%.1f}",
+        props.name, props.gcnArchName, device, props.multiProcessorCount,
+        props.clockRate / 1000, (size_t)props.maxSharedMemoryPerMultiProcessor / 1024,
+        props.warpSize, (double)mem_total / 1.0e9, (double)mem_free / 1.0e9,
+        hbm_gbps, tflops);
+    if (written < 0 || (size_t)written >= buflen) {
+        snprintf(g_err, sizeof(g_err), "probe json buffer too small");
+        return -2;
+    }
+    return 0;
+}
+
+// ---------------------------------------------------------------------------
+// hipHostMalloc-pinned staging
+// ---------------------------------------------------------------------------
+
+extern "C" void* csp_host_alloc(size_t nbytes) {
+    void* p = nullptr;
+    hipError_t e = hipHostMalloc(&p, nbytes, hipHostMallocDefault);
+    if (e != hipSuccess) {
+        set_err("hipHostMalloc", e);
+        return nullptr;
+    }
+    return p;
+}
+
+extern "C" int csp_host_free(void* p) {
+    HIP_TRY(hipHostFree(p));
+    return 0;
+}
+
+// Grow-only pooled staging buffer: electrons in one stub process (and the
+// future persistent worker) reuse a single pinned allocation.
+static std::mutex g_staging_mu;
+static void* g_staging_buf = nullptr;
+static size_t g_staging_cap = 0;
+
+extern "C" void* csp_staging_get(size_t nbytes) {
+    std::lock_guard<std::mutex> lock(g_staging_mu);
+    if (nbytes <= g_staging_cap) return g_staging_buf;
+    if (g_staging_buf) {
+        hipError_t e = hipHostFree(g_staging_buf);
+        g_staging_buf = nullptr;
+        g_staging_cap = 0;
+        if (e != hipSuccess) {
+            set_err("hipHostFree(staging)", e);
+            return nullptr;
+        }
+    }
+    void* p = nullptr;
+    hipError_t e = hipHostMalloc(&p, nbytes, hipHostMallocDefault);
+    if (e != hipSuccess) {
+        set_err("hipHostMalloc(staging)", e);
+        return nullptr;
+    }
+    g_staging_buf = p;
+    g_staging_cap = nbytes;
+    return p;
+}
+
+extern "C" int csp_staging_reset() {
+    std::lock_guard<std::mutex> lock(g_staging_mu);
+    if (g_staging_buf) HIP_TRY(hipHostFree(g_staging_buf));
+    g_staging_buf = nullptr;
+    g_staging_cap = 0;
+    return 0;
+}
+
+// Dedicated copy stream so staging copies never serialize behind the
+// caller's compute stream.
+static hipStream_t copy_stream() {
+    static hipStream_t s = nullptr;
+    static std::once_flag once;
+    std::call_once(once, [] {
+        if (hipStreamCreateWithFlags(&s, hipStreamNonBlocking) != hipSuccess)
+            s = nullptr;  // fall back to the null stream
+    });
+    return s;
+}
+
+extern "C" int csp_memcpy_d2h(void* dst_host, const void* src_dev, size_t n) {
+    hipStream_t s = copy_stream();
+    HIP_TRY(hipMemcpyAsync(dst_host, src_dev, n, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    return 0;
+}
+
+extern "C" int csp_memcpy_h2d(void* dst_dev, const void* src_host, size_t n) {
+    hipStream_t s = copy_stream();
+    HIP_TRY(hipMemcpyAsync(dst_dev, src_host, n, hipMemcpyHostToDevice, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    return 0;
+}

@@ -1,0 +1,46 @@
+"""Renderer for the remote execution stub.
+
+The reference renders its exec.py template with ``str.format`` and three
+placeholders (reference ssh.py:160-171, exec.py:12-14).  This build uses
+token substitution (``__CSP_*__``) instead so the template stays a valid,
+brace-safe Python file, and adds tokens for the GPU library path, the
+meta/timings file and staging policy.
+"""
+
+from __future__ import annotations
+
+from pathlib import Path
+from typing import Optional
+
+_TEMPLATE_PATH = Path(__file__).parent / "stub_template.py"
+
+# 64 MiB: below this a plain torch .cpu() copy is cheap; above it the
+# hipHostMalloc-pinned D2H path wins (PCIe Gen5 x16 ~63 GB/s vs pageable).
+DEFAULT_STAGING_THRESHOLD = 64 * 1024 * 1024
+
+
+def render_stub(
+    *,
+    remote_result_file: str,
+    remote_function_file: str,
+    current_remote_workdir: str,
+    remote_meta_file: str = "",
+    gpu_lib_path: str = "",
+    warmup: bool = True,
+    staging_threshold: Optional[int] = None,
+) -> str:
+    """Return the per-task stub script text."""
+    text = _TEMPLATE_PATH.read_text()
+    thr = DEFAULT_STAGING_THRESHOLD if staging_threshold is None else int(staging_threshold)
+    replacements = {
+        "__CSP_RESULT_FILE__": remote_result_file,
+        "__CSP_FUNCTION_FILE__": remote_function_file,
+        "__CSP_WORKDIR__": current_remote_workdir,
+        "__CSP_META_FILE__": remote_meta_file,
+        "__CSP_GPU_LIB__": gpu_lib_path,
+        "__CSP_WARMUP__": repr(bool(warmup)),
+        "__CSP_STAGING_THRESHOLD__": str(thr),
+    }
+    for token, value in replacements.items():
+        text = text.replace(token, value)
+    return text

@@ -1,0 +1,217 @@
+"""Remote execution stub for the MI355X SSH executor.
+
+This file is NEVER imported by the plugin.  It is read as text, the
+CSP placeholder tokens are substituted per task
+(covalent_ssh_plugin_amd/remote/stub.py), and the rendered script is
+shipped to the remote host and run there as
+``{python_path} exec_{dispatch_id}_{node_id}.py`` — the analog of the
+reference's templated exec.py (/root/reference/covalent_ssh_plugin/
+exec.py:12-46), with the same on-disk contract:
+
+* reads ``(fn, args, kwargs)`` cloudpickled into the function file,
+* writes ``(result, exception)`` pickled into the result file — exactly
+  one of the two is non-None; the process exits 0 even when the task
+  raised (errors travel in the pickle),
+* if cloudpickle is missing it writes ``(None, ImportError)`` with the
+  stdlib pickler and exits 1.
+
+MI355X additions (no reference counterpart): before the
+user function runs, a hand-written CDNA4 HIP warm-up/device-probe kernel
+(libcsp_gpu.so, ctypes) spins up the assigned GPU's clocks and records
+device facts; after it returns, CUDA-device tensors in the result are
+staged to host through hipHostMalloc-pinned memory before pickling.  A
+meta JSON with per-phase timings is written next to the result file.
+"""
+
+import json
+import os
+import sys
+import time
+from pathlib import Path
+
+RESULT_FILE = "__CSP_RESULT_FILE__"
+FUNCTION_FILE = "__CSP_FUNCTION_FILE__"
+WORKDIR = "__CSP_WORKDIR__"
+META_FILE = "__CSP_META_FILE__"
+GPU_LIB = "__CSP_GPU_LIB__"  # empty string -> no GPU library shipped
+if GPU_LIB:
+    # Resolve against the login cwd ($HOME) BEFORE the workdir chdir so
+    # ctypes.CDLL gets an absolute path.
+    GPU_LIB = os.path.abspath(os.path.expanduser(GPU_LIB))
+DO_WARMUP = bool(__CSP_WARMUP__)
+STAGING_THRESHOLD = int(__CSP_STAGING_THRESHOLD__)
+
+_t0 = time.monotonic()
+_meta = {
+    "phases_ms": {},
+    "gpu": None,
+    "staging": None,
+    "hip_visible_devices": os.environ.get("HIP_VISIBLE_DEVICES"),
+    "pid": os.getpid(),
+}
+
+
+def _mark(name, since):
+    _meta["phases_ms"][name] = round((time.monotonic() - since) * 1000.0, 3)
+
+
+def _write_meta():
+    if META_FILE:
+        try:
+            _meta["phases_ms"]["total"] = round((time.monotonic() - _t0) * 1000.0, 3)
+            Path(META_FILE).write_text(json.dumps(_meta))
+        except Exception:
+            pass
+
+
+# -- cloudpickle bootstrap (contract: reference exec.py:16-24) -------------
+try:
+    import cloudpickle as pickle
+except Exception as e:  # noqa: BLE001
+    import pickle as _stdlib_pickle
+
+    with open(RESULT_FILE, "wb") as f_out:
+        _stdlib_pickle.dump((None, e), f_out)
+    _write_meta()
+    sys.exit(1)
+
+
+def _load_gpu_lib():
+    import ctypes
+
+    lib = ctypes.CDLL(GPU_LIB)
+    lib.csp_probe_json.restype = ctypes.c_int
+    lib.csp_probe_json.argtypes = [ctypes.c_int, ctypes.c_char_p, ctypes.c_size_t]
+    lib.csp_warmup.restype = ctypes.c_int
+    lib.csp_warmup.argtypes = [ctypes.c_int, ctypes.c_int]
+    lib.csp_staging_get.restype = ctypes.c_void_p
+    lib.csp_staging_get.argtypes = [ctypes.c_size_t]
+    lib.csp_memcpy_d2h.restype = ctypes.c_int
+    lib.csp_memcpy_d2h.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_size_t]
+    lib.csp_last_error.restype = ctypes.c_char_p
+    return lib
+
+
+def _gpu_prologue():
+    """Warm up and probe the assigned MI355X before user code runs."""
+    import ctypes
+
+    t = time.monotonic()
+    lib = _load_gpu_lib()
+    buf = ctypes.create_string_buffer(8192)
+    rc = lib.csp_probe_json(0, buf, len(buf))
+    if rc != 0:
+        raise RuntimeError(
+            "csp_probe_json failed: %s" % lib.csp_last_error().decode(errors="replace")
+        )
+    _meta["gpu"] = json.loads(buf.value.decode())
+    if DO_WARMUP:
+        rc = lib.csp_warmup(0, 50)  # ~50 ms MFMA+HBM spin
+        if rc != 0:
+            raise RuntimeError(
+                "csp_warmup failed: %s" % lib.csp_last_error().decode(errors="replace")
+            )
+    _mark("gpu_prologue", t)
+    return lib
+
+
+def _stage_result(result, lib):
+    """Move CUDA-device tensors in ``result`` to host memory so the pickle
+    is portable.  Large tensors go through the hipHostMalloc-pinned
+    staging buffer (fast D2H); small ones use torch's own path."""
+    if "torch" not in sys.modules:
+        return result
+    import ctypes
+
+    import torch
+
+    stats = {"tensors": 0, "pinned_tensors": 0, "bytes": 0, "mode": "none"}
+
+    def to_host(t):
+        stats["tensors"] += 1
+        nbytes = t.numel() * t.element_size()
+        stats["bytes"] += nbytes
+        if lib is not None and nbytes >= STAGING_THRESHOLD:
+            src = t.contiguous()
+            torch.cuda.synchronize()
+            dst = lib.csp_staging_get(nbytes)
+            if dst:
+                rc = lib.csp_memcpy_d2h(
+                    ctypes.c_void_p(dst), ctypes.c_void_p(src.data_ptr()), nbytes
+                )
+                if rc == 0:
+                    raw = ctypes.string_at(dst, nbytes)
+                    host = torch.frombuffer(
+                        bytearray(raw), dtype=src.dtype
+                    ).reshape(src.shape)
+                    stats["pinned_tensors"] += 1
+                    stats["mode"] = "pinned"
+                    return host
+        stats["mode"] = stats["mode"] if stats["mode"] == "pinned" else "torch"
+        return t.cpu()
+
+    def walk(obj):
+        if isinstance(obj, torch.Tensor):
+            return to_host(obj) if obj.is_cuda else obj
+        if isinstance(obj, dict):
+            return {k: walk(v) for k, v in obj.items()}
+        if isinstance(obj, tuple):
+            vals = [walk(v) for v in obj]
+            # namedtuples construct from positional fields
+            return type(obj)(*vals) if hasattr(obj, "_fields") else tuple(vals)
+        if isinstance(obj, list):
+            return [walk(v) for v in obj]
+        return obj
+
+    staged = walk(result)
+    _meta["staging"] = stats
+    return staged
+
+
+def main():
+    with open(FUNCTION_FILE, "rb") as f_in:
+        fn, args, kwargs = pickle.load(f_in)
+
+    current_dir = os.getcwd()
+    workdir = Path(WORKDIR)
+    workdir.mkdir(parents=True, exist_ok=True)
+    os.chdir(workdir)
+
+    result = None
+    exception = None
+    gpu_lib = None
+
+    # GPU prologue runs only when a library was shipped AND a GPU slot was
+    # assigned.  A missing/broken HIP library on a box that *was* assigned
+    # a GPU is a loud error, not a silent CPU fallback.
+    if GPU_LIB and os.environ.get("HIP_VISIBLE_DEVICES") is not None:
+        try:
+            gpu_lib = _gpu_prologue()
+        except Exception as e:  # noqa: BLE001
+            exception = e
+
+    if exception is None:
+        t_fn = time.monotonic()
+        try:
+            result = fn(*args, **kwargs)
+        except Exception as e:  # noqa: BLE001
+            exception = e
+        _mark("user_fn", t_fn)
+
+    os.chdir(current_dir)
+
+    if exception is None and result is not None:
+        t_stage = time.monotonic()
+        try:
+            result = _stage_result(result, gpu_lib)
+        except Exception as e:  # noqa: BLE001
+            result, exception = None, e
+        _mark("staging", t_stage)
+
+    with open(RESULT_FILE, "wb") as f_out:
+        pickle.dump((result, exception), f_out)
+    _write_meta()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,682 @@
+"""MI355X-native Covalent SSH executor.
+
+Drop-in replacement for the reference SSHExecutor
+(/root/reference/covalent_ssh_plugin/ssh.py) with the same public
+constructor surface (reference ssh.py:75-92), the same config keys under
+``[executors.ssh]`` (reference ssh.py:39-50), and the same on-disk
+function/result pickle contract (reference ssh.py:147-150, exec.py:44-46)
+— re-architected for a remote 8×MI355X node:
+
+* **Pooled, multiplexed transport** — one OpenSSH ControlMaster per
+  (host, user, key) shared by all tasks and executor instances, instead
+  of one asyncssh connection per task (reference ssh.py:497, 585-587).
+* **Fused single-round-trip dispatch** — staging (tar on stdin), remote
+  execution, result + meta streaming and remote cleanup happen in ONE
+  ssh command, vs ~10-11 sequential round trips per electron in the
+  reference.
+* **GPU slot scheduling** — each task acquires one of the node's MI355X
+  GPUs and runs with ``HIP_VISIBLE_DEVICES`` pinned to it; slot tables
+  are shared module-level state keyed by endpoint.
+* **CDNA4 warm-up/device-probe kernel + hipHostMalloc-pinned staging**
+  in the remote stub (see remote/stub_template.py and ops/hip/).
+* Per-endpoint environment checks (python version, conda env, GPU lib
+  provisioning) are hoisted out of the per-task path and cached.
+* The reference's connection leak when the task raised (reference
+  ssh.py:581-587) does not exist here: transports are pooled, never
+  per-task, and slots release via try/finally.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import os
+import pickle as stdlib_pickle
+import shlex
+import time
+import uuid
+from pathlib import Path
+from typing import Any, Callable, Dict, List, Optional, Tuple
+
+import cloudpickle
+
+from .compat import RemoteExecutor, app_log, get_config, update_config_defaults
+from .gpu import slots as gpu_slots
+from .remote.stub import DEFAULT_STAGING_THRESHOLD, render_stub
+from .transport import (
+    CompletedCommand,
+    LocalTransport,
+    OpenSSHTransport,
+    Transport,
+    TransportConnectError,
+)
+from .transport import pool as transport_pool
+from .utils.timing import PhaseTimer, TaskRecord
+
+EXECUTOR_PLUGIN_NAME = "SSHExecutor"
+
+# Config defaults registered under [executors.ssh].  The first ten keys
+# mirror the reference exactly (reference ssh.py:39-50); the rest are
+# MI355X-native additions (SURVEY.md §5 "Config / flag system").
+_EXECUTOR_PLUGIN_DEFAULTS = {
+    "username": "",
+    "hostname": "",
+    "ssh_key_file": os.path.join(os.path.expanduser("~"), ".ssh", "id_rsa"),
+    "cache_dir": os.path.join(
+        os.environ.get("XDG_CACHE_HOME", os.path.join(os.path.expanduser("~"), ".cache")),
+        "covalent",
+    ),
+    "python_path": "python",
+    "conda_env": "",
+    "remote_cache": ".cache/covalent",
+    "run_local_on_ssh_fail": False,
+    "remote_workdir": "covalent-workdir",
+    "create_unique_workdir": False,
+    # --- MI355X-native additions -------------------------------------
+    "transport": "ssh",  # "ssh" (OpenSSH ControlMaster) | "local" (loopback)
+    "ssh_port": 22,
+    "gpu_slots": 8,  # GPUs on the remote node (8×MI355X)
+    "slots_per_gpu": 1,  # 288 GB HBM3E/GPU: 1 electron per GPU by default
+    "hip_visible_devices_policy": "roundrobin",  # "roundrobin" | "none"
+    "warmup_gpu": True,  # run the CDNA4 warm-up/probe kernel pre-task
+    "pinned_staging_threshold_bytes": DEFAULT_STAGING_THRESHOLD,
+    "batch_roundtrips": True,  # fused single-round-trip dispatch
+}
+
+update_config_defaults("executors.ssh", _EXECUTOR_PLUGIN_DEFAULTS)
+
+
+def _conf(key: str, explicit: Any, default: Any = None) -> Any:
+    """Reference resolution order: explicit ctor arg -> config -> default
+    (reference ssh.py:94-124), hardened against missing config keys."""
+    if explicit is not None and explicit != "":
+        return explicit
+    try:
+        value = get_config(f"executors.ssh.{key}")
+    except KeyError:
+        value = None
+    if value is None or value == "":
+        return default if default is not None else _EXECUTOR_PLUGIN_DEFAULTS.get(key)
+    return value
+
+
+class SSHTaskError(RuntimeError):
+    """Dispatcher-side failure of the SSH pipeline (not the user task)."""
+
+
+class SSHExecutor(RemoteExecutor):
+    """Async executor running one covalent electron per call on a remote
+    (8×MI355X) host over pooled SSH."""
+
+    def __init__(
+        self,
+        username: str = "",
+        hostname: str = "",
+        ssh_key_file: str = "",
+        cache_dir: str = "",
+        python_path: str = "",
+        conda_env: str = "",
+        remote_cache: str = "",
+        run_local_on_ssh_fail: Optional[bool] = None,
+        remote_workdir: str = "",
+        create_unique_workdir: Optional[bool] = None,
+        poll_freq: int = 15,
+        do_cleanup: bool = True,
+        retry_connect: bool = True,
+        max_connection_attempts: int = 5,
+        retry_wait_time: int = 5,
+        *,
+        transport: str = "",
+        ssh_port: Optional[int] = None,
+        gpu_slots: Optional[int] = None,
+        slots_per_gpu: Optional[int] = None,
+        hip_visible_devices_policy: str = "",
+        warmup_gpu: Optional[bool] = None,
+        pinned_staging_threshold_bytes: Optional[int] = None,
+        batch_roundtrips: Optional[bool] = None,
+        local_home: str = "",
+    ) -> None:
+        remote_cache = _conf("remote_cache", remote_cache)
+        super().__init__(poll_freq=poll_freq, remote_cache=remote_cache)
+
+        self.username = _conf("username", username, default="")
+        self.hostname = _conf("hostname", hostname, default="")
+        self.python_path = _conf("python_path", python_path)
+        self.conda_env = _conf("conda_env", conda_env, default="")
+        self.run_local_on_ssh_fail = bool(
+            _conf("run_local_on_ssh_fail", run_local_on_ssh_fail, default=False)
+        )
+        self.remote_workdir = _conf("remote_workdir", remote_workdir)
+        self.create_unique_workdir = bool(
+            _conf("create_unique_workdir", create_unique_workdir, default=False)
+        )
+        self.do_cleanup = do_cleanup
+        self.retry_connect = retry_connect
+        self.max_connection_attempts = max_connection_attempts
+        self.retry_wait_time = retry_wait_time
+
+        cache = _conf("cache_dir", cache_dir)
+        self.cache_dir = str(Path(cache).expanduser().resolve())
+        key_file = _conf("ssh_key_file", ssh_key_file)
+        self.ssh_key_file = str(Path(key_file).expanduser()) if key_file else ""
+
+        # MI355X-native knobs
+        self.transport_kind = _conf("transport", transport)
+        self.ssh_port = int(_conf("ssh_port", ssh_port))
+        self.gpu_slots = int(_conf("gpu_slots", gpu_slots))
+        self.slots_per_gpu = int(_conf("slots_per_gpu", slots_per_gpu))
+        self.hip_visible_devices_policy = _conf(
+            "hip_visible_devices_policy", hip_visible_devices_policy
+        )
+        self.warmup_gpu = bool(_conf("warmup_gpu", warmup_gpu, default=True))
+        self.pinned_staging_threshold_bytes = int(
+            _conf("pinned_staging_threshold_bytes", pinned_staging_threshold_bytes)
+        )
+        self.batch_roundtrips = bool(_conf("batch_roundtrips", batch_roundtrips, default=True))
+        self.local_home = local_home
+
+        #: most recent completed task's per-phase timing record
+        self.last_task_record: Optional[TaskRecord] = None
+        #: bounded history of task records (bench reads this)
+        self.task_records: List[TaskRecord] = []
+
+    # ------------------------------------------------------------------
+    # Endpoint identity / pooled state
+    # ------------------------------------------------------------------
+
+    def _pool_key(self) -> Tuple[str, ...]:
+        if self.transport_kind == "local":
+            return ("local", self.local_home or os.path.expanduser("~"))
+        return ("ssh", self.hostname, self.username, self.ssh_key_file, str(self.ssh_port))
+
+    def _make_transport(self) -> Transport:
+        if self.transport_kind == "local":
+            return LocalTransport(home=self.local_home or None)
+        return OpenSSHTransport(
+            hostname=self.hostname,
+            username=self.username,
+            ssh_key_file=self.ssh_key_file,
+            port=self.ssh_port,
+        )
+
+    def _slot_table(self) -> gpu_slots.SlotTable:
+        return gpu_slots.get_slot_table(
+            self._pool_key(), num_gpus=self.gpu_slots, slots_per_gpu=self.slots_per_gpu
+        )
+
+    # ------------------------------------------------------------------
+    # Failure policy (reference ssh.py:181-208)
+    # ------------------------------------------------------------------
+
+    async def _on_ssh_fail(
+        self, function: Callable, args: list, kwargs: dict, message: str
+    ) -> Any:
+        """If ``run_local_on_ssh_fail``, run the task on the dispatcher
+        host; otherwise raise (reference ssh.py:202-208)."""
+        app_log.warning("SSH dispatch failed: %s", message)
+        if self.run_local_on_ssh_fail:
+            return await asyncio.to_thread(function, *args, **kwargs)
+        raise RuntimeError(message)
+
+    # ------------------------------------------------------------------
+    # RemoteExecutor template methods (reference ssh.py:317-464)
+    # ------------------------------------------------------------------
+
+    async def _validate_credentials(self, raise_exception: bool = True) -> bool:
+        """SSH key file must exist (reference ssh.py:317-335).  The local
+        transport needs no credentials."""
+        if self.transport_kind == "local":
+            return True
+        if self.ssh_key_file and os.path.exists(self.ssh_key_file):
+            return True
+        if raise_exception:
+            raise RuntimeError(f"no SSH key file found at {self.ssh_key_file}")
+        return False
+
+    async def _client_connect(self) -> Optional[Transport]:
+        """Connect (or fetch from pool) with the reference's retry policy:
+        up to ``max_connection_attempts`` attempts spaced by
+        ``retry_wait_time`` (reference ssh.py:237-282); returns None on
+        exhaustion, raises immediately if ``retry_connect`` is False."""
+        attempts = self.max_connection_attempts if self.retry_connect else 1
+        last_error: Optional[Exception] = None
+        for attempt in range(attempts):
+            try:
+                return await transport_pool.get_transport(
+                    self._pool_key(), self._make_transport
+                )
+            except (TransportConnectError, OSError, asyncio.TimeoutError) as e:
+                last_error = e
+                if not self.retry_connect:
+                    raise
+                app_log.warning(
+                    "connect attempt %d/%d to %s failed: %s",
+                    attempt + 1,
+                    attempts,
+                    self.hostname or "local",
+                    e,
+                )
+                if attempt + 1 < attempts:
+                    await asyncio.sleep(self.retry_wait_time)
+        app_log.error("connection exhausted after %d attempts: %s", attempts, last_error)
+        return None
+
+    async def _upload_task(
+        self, transport: Transport, files: List[Tuple[str, str]]
+    ) -> None:
+        """Batched upload of the staged task files (replaces the
+        reference's two scp calls, ssh.py:360-361)."""
+        await transport.put_files(files)
+
+    def _submit_command(self, remote_script_file: str) -> str:
+        """Remote command string: ``{python_path} {script}`` with the
+        conda activation wrapper when ``conda_env`` is set (reference
+        ssh.py:377-380)."""
+        cmd = f"{self.python_path} {shlex.quote(remote_script_file)}"
+        if self.conda_env:
+            cmd = (
+                'eval "$(conda shell.bash hook)" && '
+                f"conda activate {shlex.quote(self.conda_env)} && {cmd}"
+            )
+        return cmd
+
+    async def submit_task(
+        self,
+        transport: Transport,
+        remote_script_file: str,
+        env: Optional[dict] = None,
+    ) -> CompletedCommand:
+        """Synchronous submit: awaits the remote process exit (reference
+        ssh.py:363-386)."""
+        return await transport.run(self._submit_command(remote_script_file), env=env)
+
+    async def get_status(self, transport: Transport, remote_result_file: str) -> bool:
+        """True iff the remote result file exists (the reference compares
+        ``ls`` output for equality, ssh.py:402-406)."""
+        proc = await transport.run(f"ls {shlex.quote(remote_result_file)}")
+        return proc.ok and proc.text_out().strip() == remote_result_file
+
+    async def _poll_task(
+        self, transport: Transport, remote_result_file: str, retries: int = 5
+    ) -> bool:
+        """Poll for the result file.  First check is immediate (submit is
+        synchronous, so the file normally exists already — SURVEY.md §7
+        latency note); then sleep ``poll_freq`` between at most
+        ``retries`` further checks (reference ssh.py:408-432)."""
+        for attempt in range(retries + 1):
+            if await self.get_status(transport, remote_result_file):
+                return True
+            if attempt < retries:
+                await asyncio.sleep(self.poll_freq)
+        return False
+
+    async def query_result(
+        self, transport: Transport, remote_result_file: str, local_result_file: str
+    ) -> Tuple[Any, Optional[Exception]]:
+        """Fetch and unpickle the ``(result, exception)`` 2-tuple
+        (reference ssh.py:434-458)."""
+        await transport.get_file(remote_result_file, local_result_file)
+        with open(local_result_file, "rb") as f:
+            return stdlib_pickle.load(f)
+
+    async def cancel(self, *args: Any, **kwargs: Any) -> None:
+        """Unsupported, as in the reference (ssh.py:460-464)."""
+        raise NotImplementedError("Cancellation is not supported by the SSH executor")
+
+    # ------------------------------------------------------------------
+    # Staging (reference ssh.py:126-179)
+    # ------------------------------------------------------------------
+
+    def _task_paths(self, operation_id: str) -> Dict[str, str]:
+        rc = self.remote_cache
+        return {
+            "function_local": os.path.join(self.cache_dir, f"function_{operation_id}.pkl"),
+            "script_local": os.path.join(self.cache_dir, f"exec_{operation_id}.py"),
+            "result_local": os.path.join(self.cache_dir, f"result_{operation_id}.pkl"),
+            "meta_local": os.path.join(self.cache_dir, f"meta_{operation_id}.json"),
+            "function_remote": f"{rc}/function_{operation_id}.pkl",
+            "script_remote": f"{rc}/exec_{operation_id}.py",
+            "result_remote": f"{rc}/result_{operation_id}.pkl",
+            "meta_remote": f"{rc}/meta_{operation_id}.json",
+        }
+
+    def _write_function_files(
+        self,
+        operation_id: str,
+        fn: Callable,
+        args: list,
+        kwargs: dict,
+        current_remote_workdir: str,
+        gpu_lib_path: str = "",
+    ) -> Dict[str, str]:
+        """Serialize ``(fn, args, kwargs)`` with cloudpickle and render
+        the exec stub (reference ssh.py:126-179; same file names and
+        pickle layout)."""
+        paths = self._task_paths(operation_id)
+        Path(self.cache_dir).mkdir(parents=True, exist_ok=True)
+        with open(paths["function_local"], "wb") as f:
+            cloudpickle.dump((fn, args, kwargs), f)
+        script = render_stub(
+            remote_result_file=paths["result_remote"],
+            remote_function_file=paths["function_remote"],
+            current_remote_workdir=current_remote_workdir,
+            remote_meta_file=paths["meta_remote"],
+            gpu_lib_path=gpu_lib_path,
+            warmup=self.warmup_gpu,
+            staging_threshold=self.pinned_staging_threshold_bytes,
+        )
+        Path(paths["script_local"]).write_text(script)
+        return paths
+
+    # ------------------------------------------------------------------
+    # One-time per-endpoint environment checks (hoisted out of the
+    # per-task path; reference does these per task at ssh.py:508-532)
+    # ------------------------------------------------------------------
+
+    async def _ensure_environment(self, transport: Transport) -> str:
+        """Verify python/conda once per endpoint and provision the GPU
+        library.  Returns the remote GPU library path ('' if absent)."""
+        key = self._pool_key()
+        cached = transport_pool.cached_check(key, "env")
+        if cached is not None:
+            ok, detail, gpu_lib = cached
+            if not ok:
+                raise SSHTaskError(detail)
+            return gpu_lib
+
+        # conda env existence (reference ssh.py:508-519)
+        if self.conda_env:
+            proc = await transport.run(
+                f'eval "$(conda shell.bash hook)" && conda env list | grep {shlex.quote(self.conda_env)}'
+            )
+            if not proc.ok:
+                detail = f"conda environment {self.conda_env!r} not found on {transport.endpoint}"
+                transport_pool.store_check(key, "env", (False, detail, ""))
+                raise SSHTaskError(detail)
+
+        # python3 sanity (reference ssh.py:521-524) + remote cache dir +
+        # GPU presence, all in one round trip
+        proc = await transport.run(
+            self._wrap_conda(f"{self.python_path} --version")
+            + f" && mkdir -p {shlex.quote(self.remote_cache)}"
+            + " && { test -e /dev/kfd && echo CSP_HAS_GPU || true; }"
+        )
+        if not proc.ok or "3" not in (proc.text_out() + proc.text_err()):
+            detail = (
+                f"no python3 at {self.python_path!r} on {transport.endpoint}: "
+                f"{proc.text_err().strip()}"
+            )
+            transport_pool.store_check(key, "env", (False, detail, ""))
+            raise SSHTaskError(detail)
+
+        # Only provision the CDNA4 library (and later inject GPU slots)
+        # when the endpoint actually has an AMD GPU stack.
+        has_gpu = "CSP_HAS_GPU" in proc.text_out()
+        gpu_lib = await self._provision_gpu_lib(transport) if has_gpu else ""
+        transport_pool.store_check(key, "env", (True, "", gpu_lib))
+        return gpu_lib
+
+    def _wrap_conda(self, cmd: str) -> str:
+        if self.conda_env:
+            return (
+                'eval "$(conda shell.bash hook)" && '
+                f"conda activate {shlex.quote(self.conda_env)} && {cmd}"
+            )
+        return cmd
+
+    async def _provision_gpu_lib(self, transport: Transport) -> str:
+        """Ship the in-tree libcsp_gpu.so (CDNA4 probe/warm-up + pinned
+        staging) to the endpoint once, content-addressed."""
+        from .gpu.probe import local_gpu_lib_path
+
+        local_lib = local_gpu_lib_path()
+        if not local_lib:
+            return ""
+        import hashlib
+
+        digest = hashlib.sha256(Path(local_lib).read_bytes()).hexdigest()[:12]
+        remote_lib = f"{self.remote_cache}/lib/csp_gpu-{digest}.so"
+        probe = await transport.run(f"test -f {shlex.quote(remote_lib)}")
+        if not probe.ok:
+            await transport.put_files([(local_lib, remote_lib)])
+        return remote_lib
+
+    # ------------------------------------------------------------------
+    # Fused single-round-trip dispatch
+    # ------------------------------------------------------------------
+
+    @staticmethod
+    def _split_stream(
+        stdout: bytes, s_result: bytes, s_meta: bytes
+    ) -> Tuple[bytes, Optional[bytes], Optional[bytes]]:
+        """Split the fused command's stdout into (task_stdout,
+        result_bytes, meta_bytes) at the per-task sentinels."""
+        idx = stdout.find(s_result)
+        if idx < 0:
+            return stdout, None, None
+        task_out = stdout[:idx]
+        rest = stdout[idx + len(s_result):]
+        midx = rest.find(s_meta)
+        if midx < 0:
+            return task_out, rest, None
+        return task_out, rest[:midx], rest[midx + len(s_meta):]
+
+    async def _dispatch_fused(
+        self,
+        transport: Transport,
+        paths: Dict[str, str],
+        env: Optional[dict],
+    ) -> Tuple[CompletedCommand, Optional[bytes], Optional[bytes]]:
+        """Stage + execute + fetch + clean in ONE transport round trip.
+
+        stdin carries the staged files as a tar stream; stdout carries the
+        task's own stdout, then sentinel-delimited result and meta bytes.
+        """
+        from .transport.base import make_tar_stream
+
+        files = [
+            (paths["function_local"], paths["function_remote"]),
+            (paths["script_local"], paths["script_remote"]),
+        ]
+        tar_bytes, base = make_tar_stream(files)
+        untar = "tar -xf - -C /" if base == "/" else "tar -xf -"
+
+        token = uuid.uuid4().hex
+        s_result = f"\n--CSP-RESULT-{token}--\n"
+        s_meta = f"\n--CSP-META-{token}--\n"
+        q = shlex.quote
+        submit = self._submit_command(paths["script_remote"])
+        rm_files = " ".join(
+            q(paths[k]) for k in ("function_remote", "script_remote", "result_remote", "meta_remote")
+        )
+        cleanup = f"rm -f {rm_files}; " if self.do_cleanup else ""
+        cmd = (
+            f"mkdir -p {q(self.remote_cache)} && {untar} && "
+            f"{submit}; _csp_rc=$?; "
+            f"if [ -f {q(paths['result_remote'])} ]; then "
+            f"printf '%s' {q(s_result)}; cat {q(paths['result_remote'])}; "
+            f"printf '%s' {q(s_meta)}; cat {q(paths['meta_remote'])} 2>/dev/null || true; "
+            f"fi; {cleanup}exit $_csp_rc"
+        )
+        proc = await transport.run(cmd, input_data=tar_bytes, env=env)
+        task_out, result_bytes, meta_bytes = self._split_stream(
+            proc.stdout, s_result.encode(), s_meta.encode()
+        )
+        if task_out:
+            app_log.debug("task stdout: %s", task_out.decode(errors="replace"))
+        return (
+            CompletedCommand(proc.returncode, task_out, proc.stderr),
+            result_bytes,
+            meta_bytes,
+        )
+
+    # ------------------------------------------------------------------
+    # run() — the dispatcher-invoked entry point (reference ssh.py:466-591)
+    # ------------------------------------------------------------------
+
+    async def run(
+        self,
+        function: Callable,
+        args: list,
+        kwargs: dict,
+        task_metadata: Optional[dict] = None,
+    ) -> Any:
+        task_metadata = task_metadata or {}
+        dispatch_id = task_metadata.get("dispatch_id", "dispatch")
+        node_id = task_metadata.get("node_id", 0)
+        operation_id = f"{dispatch_id}_{node_id}"
+
+        if self.create_unique_workdir:
+            current_remote_workdir = (
+                f"{self.remote_workdir}/{dispatch_id}/node_{node_id}"
+            )
+        else:
+            current_remote_workdir = self.remote_workdir
+
+        timer = PhaseTimer()
+        record = TaskRecord(operation_id=operation_id)
+
+        with timer.phase("validate"):
+            await self._validate_credentials()
+
+        with timer.phase("connect"):
+            try:
+                transport = await self._client_connect()
+            except (TransportConnectError, OSError) as e:
+                return await self._on_ssh_fail(
+                    function, args, kwargs, f"Could not connect to {self.hostname}: {e}"
+                )
+        if transport is None:
+            return await self._on_ssh_fail(
+                function,
+                args,
+                kwargs,
+                f"Could not connect to {self.hostname} after "
+                f"{self.max_connection_attempts} attempts",
+            )
+
+        with timer.phase("env_checks"):
+            try:
+                gpu_lib = await self._ensure_environment(transport)
+            except SSHTaskError as e:
+                return await self._on_ssh_fail(function, args, kwargs, str(e))
+
+        slot = None
+        use_slots = self.hip_visible_devices_policy == "roundrobin" and gpu_lib
+        try:
+            if use_slots:
+                with timer.phase("slot_wait"):
+                    slot = await self._slot_table().acquire()
+                env = slot.env()
+                record.gpu_id = slot.gpu_id
+            else:
+                env = None
+
+            with timer.phase("stage"):
+                paths = self._write_function_files(
+                    operation_id, function, args, kwargs, current_remote_workdir, gpu_lib
+                )
+
+            result: Any = None
+            exception: Optional[Exception] = None
+
+            if self.batch_roundtrips:
+                with timer.phase("dispatch"):
+                    proc, result_bytes, meta_bytes = await self._dispatch_fused(
+                        transport, paths, env
+                    )
+                if proc.returncode != 0 or result_bytes is None:
+                    message = (
+                        f"remote task {operation_id} failed "
+                        f"(rc={proc.returncode}): {proc.text_err().strip()}"
+                    )
+                    self._cleanup_local(paths)
+                    return await self._on_ssh_fail(function, args, kwargs, message)
+                with timer.phase("fetch"):
+                    result, exception = stdlib_pickle.loads(result_bytes)
+                    if meta_bytes:
+                        record.load_meta(meta_bytes)
+                    if self.do_cleanup is False:
+                        Path(paths["result_local"]).write_bytes(result_bytes)
+            else:
+                # Template path: discrete upload/submit/poll/fetch/cleanup
+                # round trips (reference §3.1 flow).
+                with timer.phase("upload"):
+                    await self._upload_task(
+                        transport,
+                        [
+                            (paths["function_local"], paths["function_remote"]),
+                            (paths["script_local"], paths["script_remote"]),
+                        ],
+                    )
+                with timer.phase("dispatch"):
+                    proc = await self.submit_task(transport, paths["script_remote"], env=env)
+                if proc.returncode != 0:
+                    self._cleanup_local(paths)
+                    return await self._on_ssh_fail(
+                        function,
+                        args,
+                        kwargs,
+                        f"remote task {operation_id} failed "
+                        f"(rc={proc.returncode}): {proc.text_err().strip()}",
+                    )
+                with timer.phase("poll"):
+                    ready = await self._poll_task(transport, paths["result_remote"])
+                if not ready:
+                    self._cleanup_local(paths)
+                    return await self._on_ssh_fail(
+                        function, args, kwargs, f"result for {operation_id} never appeared"
+                    )
+                with timer.phase("fetch"):
+                    result, exception = await self.query_result(
+                        transport, paths["result_remote"], paths["result_local"]
+                    )
+                    try:
+                        await transport.get_file(paths["meta_remote"], paths["meta_local"])
+                        record.load_meta_file(paths["meta_local"])
+                    except Exception:
+                        pass
+                if self.do_cleanup:
+                    with timer.phase("cleanup"):
+                        await transport.run(
+                            "rm -f "
+                            + " ".join(
+                                shlex.quote(paths[k])
+                                for k in (
+                                    "function_remote",
+                                    "script_remote",
+                                    "result_remote",
+                                    "meta_remote",
+                                )
+                            )
+                        )
+        finally:
+            if slot is not None:
+                await slot.release()
+
+        if self.do_cleanup:
+            self._cleanup_local(paths)
+
+        record.phases = timer.snapshot()
+        record.total_s = timer.total()
+        self.last_task_record = record
+        self.task_records.append(record)
+        if len(self.task_records) > 10000:
+            del self.task_records[: len(self.task_records) // 2]
+
+        if exception is not None:
+            raise exception
+        return result
+
+    def _cleanup_local(self, paths: Dict[str, str]) -> None:
+        for key in ("function_local", "script_local", "result_local", "meta_local"):
+            try:
+                os.remove(paths[key])
+            except OSError:
+                pass
+
+    # Convenience for closing pooled transports (e.g. at interpreter exit
+    # or between tests).  The reference closes per-task; pooled transports
+    # outlive tasks by design.
+    @staticmethod
+    async def close_pool() -> None:
+        await transport_pool.close_all()

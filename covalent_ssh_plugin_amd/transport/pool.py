@@ -1,0 +1,78 @@
+"""Module-level keyed transport pool.
+
+The reference opens and closes one SSH connection per task (reference
+ssh.py:497, 585-587) and leaks it when the task raised (ssh.py:581-583 —
+bug noted in SURVEY.md §3.1).  Here, all executor instances that target
+the same (hostname, username, key, port) share ONE pooled transport; the
+pool also hosts the per-endpoint one-time environment checks (python
+version, conda env existence) so they are paid once per endpoint, not
+once per task (hoisted out of reference ssh.py:508-524).
+
+The pool is keyed module-level state on purpose: multiple SSHExecutor
+instances pointing at the same host must share connection + GPU-slot
+accounting (SURVEY.md §7 "Hard parts").
+"""
+
+from __future__ import annotations
+
+import asyncio
+from typing import Callable, Dict, Optional, Tuple
+
+from .base import Transport
+
+PoolKey = Tuple[str, ...]
+
+_pool: Dict[PoolKey, Transport] = {}
+_pool_lock: Optional[asyncio.Lock] = None
+# One-time environment check results, keyed by (pool_key, check_name).
+_env_checks: Dict[Tuple[PoolKey, str], object] = {}
+
+
+def _lock() -> asyncio.Lock:
+    # Created lazily so the lock binds to the running loop.
+    global _pool_lock
+    if _pool_lock is None:
+        _pool_lock = asyncio.Lock()
+    return _pool_lock
+
+
+async def get_transport(key: PoolKey, factory: Callable[[], Transport]) -> Transport:
+    """Return the pooled, connected transport for ``key``, creating it
+    with ``factory`` (and connecting it) on first use.
+
+    If the pooled transport has dropped its connection, reconnect it.
+    """
+    async with _lock():
+        transport = _pool.get(key)
+        if transport is None:
+            transport = factory()
+            _pool[key] = transport
+    if not transport.is_connected:
+        await transport.connect()
+    return transport
+
+
+def cached_check(key: PoolKey, name: str):
+    return _env_checks.get((key, name))
+
+
+def store_check(key: PoolKey, name: str, value: object) -> None:
+    _env_checks[(key, name)] = value
+
+
+async def close_all() -> None:
+    async with _lock():
+        transports = list(_pool.values())
+        _pool.clear()
+        _env_checks.clear()
+    for t in transports:
+        try:
+            await t.close()
+        except Exception:
+            pass
+
+
+def reset() -> None:
+    """Synchronous test hook: forget pooled state without closing."""
+    _pool.clear()
+    _env_checks.clear()
