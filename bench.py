@@ -1,0 +1,209 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: electron dispatch throughput + p50 round-trip
+latency through the MI355X-native SSH executor (BASELINE.json metric).
+
+One "step" = one complete no-op electron round trip through the full
+executor pipeline: cloudpickle staging, transport round trip, remote
+stub process spawn with the task's GPU slot pinned via
+HIP_VISIBLE_DEVICES, result pickle return, cleanup.  With N ranks
+(one per GPU, launched by torch.distributed.run), each rank drives its
+own GPU slot concurrently; the whole-job metric is electrons/sec
+aggregated over all ranks, timed over the max across ranks.
+
+No sshd exists in the benchmark image, so the measured transport is the
+loopback transport (identical semantics and process structure to the
+SSH path minus the network hop) — stated in the "data" field.  The
+reference publishes no numbers: vs_baseline is null.
+
+Usage:  python bench.py [--gpus N] [--steps K] [--warmup W]
+        [--config noop|mm|staging|rccl]
+"""
+
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import os
+import statistics
+import sys
+import tempfile
+import time
+
+REPO_ROOT = os.path.dirname(os.path.abspath(__file__))
+if REPO_ROOT not in sys.path:
+    sys.path.insert(0, REPO_ROOT)
+
+
+def build_electron(config: str):
+    """Electron bodies are defined inside a factory so cloudpickle
+    serializes them by value."""
+    if config == "noop":
+
+        def noop():
+            return 0
+
+        return noop, [], {}
+
+    if config == "mm":
+        # BASELINE config 2: single torch.mm(4096x4096, bf16) on one MI355X
+        def mm_electron(n=4096):
+            import torch
+
+            a = torch.randn(n, n, device="cuda", dtype=torch.bfloat16)
+            b = torch.randn(n, n, device="cuda", dtype=torch.bfloat16)
+            c = a @ b
+            torch.cuda.synchronize()
+            return float(c.float().mean().item())
+
+        return mm_electron, [], {}
+
+    if config == "staging":
+        # BASELINE config 4: 1 GB tensor return through pinned staging
+        def staging_electron(nbytes=1 << 30):
+            import torch
+
+            t = torch.ones(nbytes // 2, device="cuda", dtype=torch.bfloat16)
+            torch.cuda.synchronize()
+            return t
+
+        return staging_electron, [], {}
+
+    if config == "rccl":
+        from covalent_ssh_plugin_amd.parallel.rccl_smoke import (
+            make_rccl_smoke_electron,
+        )
+
+        import torch
+
+        world = torch.cuda.device_count() if torch.cuda.is_available() else 2
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        return (
+            make_rccl_smoke_electron(),
+            [],
+            {"world_size": max(1, world), "backend": backend},
+        )
+
+    raise SystemExit(f"unknown --config {config}")
+
+
+def main() -> None:
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=32)
+    parser.add_argument("--warmup", type=int, default=8)
+    parser.add_argument("--config", default="noop")
+    args = parser.parse_args()
+
+    import torch
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    distributed = world_size > 1
+
+    has_cuda = torch.cuda.is_available()
+    if distributed:
+        import torch.distributed as dist
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        backend = "nccl" if has_cuda else "gloo"
+        dist.init_process_group(backend=backend)
+        if has_cuda:
+            torch.cuda.set_device(local_rank)
+
+    def barrier_sync():
+        if distributed:
+            import torch.distributed as dist
+
+            dist.barrier()
+        if has_cuda:
+            torch.cuda.synchronize()
+
+    from covalent_ssh_plugin_amd import SSHExecutor
+
+    fn, fargs, fkwargs = build_electron(args.config)
+
+    async def run_bench() -> dict:
+        with tempfile.TemporaryDirectory() as home, tempfile.TemporaryDirectory() as cache:
+            ex = SSHExecutor(
+                transport="local",
+                local_home=home,
+                cache_dir=cache,
+                python_path=sys.executable,
+                hip_visible_devices_policy="fixed" if has_cuda else "none",
+                fixed_gpu=local_rank,
+                gpu_slots=max(1, torch.cuda.device_count()) if has_cuda else 1,
+                # dispatch-throughput metric: slot pinning yes, per-task
+                # clock warm-up no (measured separately by --config mm)
+                warmup_gpu=args.config in ("mm", "staging"),
+                pinned_staging_threshold_bytes=1 << 20,
+            )
+
+            async def one_step() -> float:
+                t0 = time.perf_counter()
+                await ex.execute(
+                    fn, list(fargs), dict(fkwargs),
+                    dispatch_id=f"bench{rank}", node_id=0,
+                )
+                return time.perf_counter() - t0
+
+            for _ in range(args.warmup):
+                await one_step()
+
+            barrier_sync()
+            t_start = time.perf_counter()
+            lat = [await one_step() for _ in range(args.steps)]
+            barrier_sync()
+            elapsed = time.perf_counter() - t_start
+            await SSHExecutor.close_pool()
+            return {"elapsed": elapsed, "latencies": lat}
+
+    out = asyncio.run(run_bench())
+    elapsed = out["elapsed"]
+
+    if distributed:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device="cuda" if has_cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    total_electrons = args.steps * world_size
+    eps = total_electrons / elapsed
+    p50_ms = statistics.median(out["latencies"]) * 1000.0
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        line = {
+            "metric": "electrons_per_sec",
+            "value": round(eps, 3),
+            "unit": "electrons/s",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if args.config in ("mm", "staging") else "n/a",
+            "data": "synthetic no-op electrons, loopback transport (no sshd in image)",
+            "p50_ms": round(p50_ms, 3),
+            "config": {
+                "model": f"{args.config}-electron-dispatch",
+                "global_batch": world_size,
+                "seq_len": 0,
+                "parallelism": f"slots{world_size}",
+            },
+        }
+        print(json.dumps(line))
+
+    if distributed:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

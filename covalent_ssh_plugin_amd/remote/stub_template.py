@@ -125,6 +125,13 @@ def _stage_result(result, lib):
 
     import torch
 
+    if lib is None and GPU_LIB and os.path.exists(GPU_LIB):
+        # warm-up was skipped but staging still wants the pinned path
+        try:
+            lib = _load_gpu_lib()
+        except Exception:  # noqa: BLE001
+            lib = None
+
     stats = {"tensors": 0, "pinned_tensors": 0, "bytes": 0, "mode": "none"}
 
     def to_host(t):
@@ -140,10 +147,13 @@ def _stage_result(result, lib):
                     ctypes.c_void_p(dst), ctypes.c_void_p(src.data_ptr()), nbytes
                 )
                 if rc == 0:
-                    raw = ctypes.string_at(dst, nbytes)
-                    host = torch.frombuffer(
-                        bytearray(raw), dtype=src.dtype
-                    ).reshape(src.shape)
+                    # zero-copy view of the pooled pinned buffer; the
+                    # result pickle copies out of it, and the pool
+                    # outlives this process's single dump
+                    view = (ctypes.c_char * nbytes).from_address(dst)
+                    host = torch.frombuffer(view, dtype=src.dtype).reshape(
+                        src.shape
+                    )
                     stats["pinned_tensors"] += 1
                     stats["mode"] = "pinned"
                     return host
@@ -181,14 +191,16 @@ def main():
     exception = None
     gpu_lib = None
 
-    # GPU prologue runs only when a library was shipped AND a GPU slot was
-    # assigned.  A missing/broken HIP library on a box that *was* assigned
-    # a GPU is a loud error, not a silent CPU fallback.
+    # GPU prologue runs only when a library was shipped, warm-up was
+    # requested AND a GPU slot was assigned.  A missing/broken HIP
+    # library on a box that *was* assigned a GPU is a loud error, not a
+    # silent CPU fallback.
     if GPU_LIB and os.environ.get("HIP_VISIBLE_DEVICES") is not None:
-        try:
-            gpu_lib = _gpu_prologue()
-        except Exception as e:  # noqa: BLE001
-            exception = e
+        if DO_WARMUP:
+            try:
+                gpu_lib = _gpu_prologue()
+            except Exception as e:  # noqa: BLE001
+                exception = e
 
     if exception is None:
         t_fn = time.monotonic()

@@ -133,6 +133,7 @@ class SSHExecutor(RemoteExecutor):
         warmup_gpu: Optional[bool] = None,
         pinned_staging_threshold_bytes: Optional[int] = None,
         batch_roundtrips: Optional[bool] = None,
+        fixed_gpu: int = 0,
         local_home: str = "",
     ) -> None:
         remote_cache = _conf("remote_cache", remote_cache)
@@ -172,6 +173,7 @@ class SSHExecutor(RemoteExecutor):
             _conf("pinned_staging_threshold_bytes", pinned_staging_threshold_bytes)
         )
         self.batch_roundtrips = bool(_conf("batch_roundtrips", batch_roundtrips, default=True))
+        self.fixed_gpu = int(fixed_gpu)
         self.local_home = local_home
 
         #: most recent completed task's per-phase timing record
@@ -561,13 +563,19 @@ class SSHExecutor(RemoteExecutor):
                 return await self._on_ssh_fail(function, args, kwargs, str(e))
 
         slot = None
-        use_slots = self.hip_visible_devices_policy == "roundrobin" and gpu_lib
+        policy = self.hip_visible_devices_policy if gpu_lib else "none"
         try:
-            if use_slots:
+            if policy == "roundrobin":
                 with timer.phase("slot_wait"):
                     slot = await self._slot_table().acquire()
                 env = slot.env()
                 record.gpu_id = slot.gpu_id
+            elif policy == "fixed":
+                env = {
+                    "HIP_VISIBLE_DEVICES": str(self.fixed_gpu),
+                    "ROCR_VISIBLE_DEVICES": str(self.fixed_gpu),
+                }
+                record.gpu_id = self.fixed_gpu
             else:
                 env = None
 
