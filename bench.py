@@ -3,8 +3,9 @@
 latency through the MI355X-native SSH executor (BASELINE.json metric).
 
 One "step" = one complete no-op electron round trip through the full
-executor pipeline: cloudpickle staging, transport round trip, remote
-stub process spawn with the task's GPU slot pinned via
+executor pipeline: cloudpickle staging, transport round trip to the
+per-GPU-slot persistent worker (or a freshly spawned stub with
+--config noop-stub), GPU slot pinning via CSP_GPU_SLOT ->
 HIP_VISIBLE_DEVICES, result pickle return, cleanup.  With N ranks
 (one per GPU, launched by torch.distributed.run), each rank drives its
 own GPU slot concurrently; the whole-job metric is electrons/sec
@@ -38,7 +39,7 @@ if REPO_ROOT not in sys.path:
 def build_electron(config: str):
     """Electron bodies are defined inside a factory so cloudpickle
     serializes them by value."""
-    if config == "noop":
+    if config in ("noop", "noop-stub"):
 
         def noop():
             return 0
@@ -138,6 +139,9 @@ def main() -> None:
                 # clock warm-up no (measured separately by --config mm)
                 warmup_gpu=args.config in ("mm", "staging"),
                 pinned_staging_threshold_bytes=1 << 20,
+                # warm worker per GPU slot (the production dispatch path);
+                # "noop-stub" measures the classic spawn-per-task stub
+                persistent_workers=args.config != "noop-stub",
             )
 
             async def one_step() -> float:
@@ -188,7 +192,10 @@ def main() -> None:
             "scaling": "weak",
             "vs_baseline": None,
             "dtype": "bf16" if args.config in ("mm", "staging") else "n/a",
-            "data": "synthetic no-op electrons, loopback transport (no sshd in image)",
+            "data": (
+                "synthetic electrons, loopback transport (no sshd in image), "
+                + ("persistent-worker dispatch" if args.config != "noop-stub" else "spawn-per-task stub dispatch")
+            ),
             "p50_ms": round(p50_ms, 3),
             "config": {
                 "model": f"{args.config}-electron-dispatch",

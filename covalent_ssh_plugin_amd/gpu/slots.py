@@ -34,10 +34,10 @@ class Slot:
     _released: bool = False
 
     def env(self) -> Dict[str, str]:
-        return {
-            "HIP_VISIBLE_DEVICES": str(self.gpu_id),
-            "ROCR_VISIBLE_DEVICES": str(self.gpu_id),
-        }
+        # The stub/worker resolves CSP_GPU_SLOT to HIP_VISIBLE_DEVICES
+        # *within* the ambient visibility list, so slot pinning composes
+        # with container/pod GPU isolation instead of overriding it.
+        return {"CSP_GPU_SLOT": str(self.gpu_id)}
 
     async def release(self) -> None:
         if not self._released:
@@ -64,11 +64,16 @@ class SlotTable:
         self._free: List[int] = [slots_per_gpu] * num_gpus
         self._order: List[int] = list(range(num_gpus))
         self._cond: Optional[asyncio.Condition] = None
+        self._cond_loop = None
         self._in_use = 0
 
     def _condition(self) -> asyncio.Condition:
-        if self._cond is None:
+        # bound to the running loop; recreated when a fresh asyncio.run()
+        # loop appears (the previous loop is closed -> no live waiters)
+        loop = asyncio.get_running_loop()
+        if self._cond is None or self._cond_loop is not loop:
             self._cond = asyncio.Condition()
+            self._cond_loop = loop
         return self._cond
 
     @property

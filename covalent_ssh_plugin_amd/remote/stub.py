@@ -13,6 +13,7 @@ from pathlib import Path
 from typing import Optional
 
 _TEMPLATE_PATH = Path(__file__).parent / "stub_template.py"
+_WORKER_TEMPLATE_PATH = Path(__file__).parent / "worker_template.py"
 
 # 64 MiB: below this a plain torch .cpu() copy is cheap; above it the
 # hipHostMalloc-pinned D2H path wins (PCIe Gen5 x16 ~63 GB/s vs pageable).
@@ -37,6 +38,26 @@ def render_stub(
         "__CSP_FUNCTION_FILE__": remote_function_file,
         "__CSP_WORKDIR__": current_remote_workdir,
         "__CSP_META_FILE__": remote_meta_file,
+        "__CSP_GPU_LIB__": gpu_lib_path,
+        "__CSP_WARMUP__": repr(bool(warmup)),
+        "__CSP_STAGING_THRESHOLD__": str(thr),
+    }
+    for token, value in replacements.items():
+        text = text.replace(token, value)
+    return text
+
+
+def render_worker(
+    *,
+    gpu_lib_path: str = "",
+    warmup: bool = True,
+    staging_threshold: Optional[int] = None,
+) -> str:
+    """Return the persistent worker script text (one per endpoint; the
+    GPU slot arrives via the CSP_GPU_SLOT env var at launch)."""
+    text = _WORKER_TEMPLATE_PATH.read_text()
+    thr = DEFAULT_STAGING_THRESHOLD if staging_threshold is None else int(staging_threshold)
+    replacements = {
         "__CSP_GPU_LIB__": gpu_lib_path,
         "__CSP_WARMUP__": repr(bool(warmup)),
         "__CSP_STAGING_THRESHOLD__": str(thr),

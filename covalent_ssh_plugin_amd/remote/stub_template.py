@@ -38,6 +38,26 @@ if GPU_LIB:
     # Resolve against the login cwd ($HOME) BEFORE the workdir chdir so
     # ctypes.CDLL gets an absolute path.
     GPU_LIB = os.path.abspath(os.path.expanduser(GPU_LIB))
+
+
+def _resolve_gpu_slot():
+    """Map the executor's abstract GPU slot (CSP_GPU_SLOT) to
+    HIP_VISIBLE_DEVICES, selecting WITHIN any ambient visibility list so
+    slot pinning composes with container/pod GPU isolation instead of
+    clobbering it.  Must run before any HIP/torch initialization."""
+    slot = os.environ.get("CSP_GPU_SLOT")
+    if slot is None:
+        return None
+    ambient = os.environ.get("HIP_VISIBLE_DEVICES") or ""
+    ids = [x for x in ambient.split(",") if x.strip() != ""]
+    if ids:
+        os.environ["HIP_VISIBLE_DEVICES"] = ids[int(slot) % len(ids)]
+    else:
+        os.environ["HIP_VISIBLE_DEVICES"] = slot
+    return slot
+
+
+GPU_SLOT = _resolve_gpu_slot()
 DO_WARMUP = bool(__CSP_WARMUP__)
 STAGING_THRESHOLD = int(__CSP_STAGING_THRESHOLD__)
 
@@ -46,6 +66,7 @@ _meta = {
     "phases_ms": {},
     "gpu": None,
     "staging": None,
+    "gpu_slot": GPU_SLOT,
     "hip_visible_devices": os.environ.get("HIP_VISIBLE_DEVICES"),
     "pid": os.getpid(),
 }
@@ -195,7 +216,7 @@ def main():
     # requested AND a GPU slot was assigned.  A missing/broken HIP
     # library on a box that *was* assigned a GPU is a loud error, not a
     # silent CPU fallback.
-    if GPU_LIB and os.environ.get("HIP_VISIBLE_DEVICES") is not None:
+    if GPU_LIB and GPU_SLOT is not None:
         if DO_WARMUP:
             try:
                 gpu_lib = _gpu_prologue()

@@ -1,0 +1,265 @@
+"""Persistent remote worker for the MI355X SSH executor.
+
+Like stub_template.py this file is NEVER imported by the plugin: it is
+rendered (CSP token substitution) and shipped to the remote host once
+per endpoint (content-addressed), then launched once per GPU slot over a
+long-lived transport channel.  Where the classic stub pays a fresh
+python + HIP-runtime start per electron (the reference's architecture,
+/root/reference/covalent_ssh_plugin/exec.py), a worker pays it once:
+
+  * starts, optionally runs the CDNA4 warm-up/device-probe prologue on
+    its pinned GPU (HIP_VISIBLE_DEVICES is set by the launcher),
+  * then serves electrons over a length-framed binary protocol on
+    stdin/stdout: each request carries the cloudpickled
+    ``(fn, args, kwargs)`` and a workdir; each reply carries the pickled
+    ``(result, exception)`` 2-tuple (same payload contract as the
+    result file) plus a meta JSON dict.
+
+Protocol (all frames: 4-byte big-endian length + payload):
+  request  = pickle dict {"op_id", "workdir", "function_blob"}
+  response = pickle tuple (result_blob: bytes, meta: dict)
+  a zero-length request frame means: shut down.
+
+The worker's REAL stdout is reserved for the protocol; fd 1 is
+re-pointed at stderr before any user code runs, so user prints cannot
+corrupt frames.
+"""
+
+import io
+import json
+import os
+import struct
+import sys
+import time
+
+GPU_LIB = "__CSP_GPU_LIB__"
+DO_WARMUP = bool(__CSP_WARMUP__)
+STAGING_THRESHOLD = int(__CSP_STAGING_THRESHOLD__)
+
+if GPU_LIB:
+    GPU_LIB = os.path.abspath(os.path.expanduser(GPU_LIB))
+
+
+def _resolve_gpu_slot():
+    """Map CSP_GPU_SLOT to HIP_VISIBLE_DEVICES within the ambient
+    visibility list (composes with pod GPU isolation).  Must run before
+    any HIP/torch initialization."""
+    slot = os.environ.get("CSP_GPU_SLOT")
+    if slot is None:
+        return None
+    ambient = os.environ.get("HIP_VISIBLE_DEVICES") or ""
+    ids = [x for x in ambient.split(",") if x.strip() != ""]
+    if ids:
+        os.environ["HIP_VISIBLE_DEVICES"] = ids[int(slot) % len(ids)]
+    else:
+        os.environ["HIP_VISIBLE_DEVICES"] = slot
+    return slot
+
+
+GPU_SLOT = _resolve_gpu_slot()
+
+# --- claim the protocol channel, push user stdout to stderr ---------------
+_proto_fd = os.dup(1)
+os.dup2(2, 1)
+sys.stdout = sys.stderr
+
+import cloudpickle as pickle  # noqa: E402  (after fd surgery on purpose)
+
+_gpu_lib = None
+_gpu_info = None
+
+
+def _load_gpu_lib():
+    import ctypes
+
+    lib = ctypes.CDLL(GPU_LIB)
+    lib.csp_probe_json.restype = ctypes.c_int
+    lib.csp_probe_json.argtypes = [ctypes.c_int, ctypes.c_char_p, ctypes.c_size_t]
+    lib.csp_warmup.restype = ctypes.c_int
+    lib.csp_warmup.argtypes = [ctypes.c_int, ctypes.c_int]
+    lib.csp_staging_get.restype = ctypes.c_void_p
+    lib.csp_staging_get.argtypes = [ctypes.c_size_t]
+    lib.csp_memcpy_d2h.restype = ctypes.c_int
+    lib.csp_memcpy_d2h.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_size_t]
+    lib.csp_last_error.restype = ctypes.c_char_p
+    return lib
+
+
+def _prologue():
+    """One-time GPU warm-up + probe for this worker's pinned device."""
+    global _gpu_lib, _gpu_info
+    import ctypes
+
+    _gpu_lib = _load_gpu_lib()
+    buf = ctypes.create_string_buffer(8192)
+    rc = _gpu_lib.csp_probe_json(0, buf, len(buf))
+    if rc != 0:
+        raise RuntimeError(
+            "csp_probe_json failed: %s"
+            % _gpu_lib.csp_last_error().decode(errors="replace")
+        )
+    _gpu_info = json.loads(buf.value.decode())
+    if DO_WARMUP:
+        rc = _gpu_lib.csp_warmup(0, 50)
+        if rc != 0:
+            raise RuntimeError(
+                "csp_warmup failed: %s"
+                % _gpu_lib.csp_last_error().decode(errors="replace")
+            )
+
+
+def _stage_result(result, stats):
+    """Pinned-staging of CUDA tensors (same policy as the stub)."""
+    if "torch" not in sys.modules:
+        return result
+    import ctypes
+
+    import torch
+
+    lib = _gpu_lib
+    if lib is None and GPU_LIB and os.path.exists(GPU_LIB):
+        try:
+            lib = _load_gpu_lib()
+        except Exception:  # noqa: BLE001
+            lib = None
+
+    def to_host(t):
+        stats["tensors"] += 1
+        nbytes = t.numel() * t.element_size()
+        stats["bytes"] += nbytes
+        if lib is not None and nbytes >= STAGING_THRESHOLD:
+            src = t.contiguous()
+            torch.cuda.synchronize()
+            dst = lib.csp_staging_get(nbytes)
+            if dst:
+                rc = lib.csp_memcpy_d2h(
+                    ctypes.c_void_p(dst), ctypes.c_void_p(src.data_ptr()), nbytes
+                )
+                if rc == 0:
+                    view = (ctypes.c_char * nbytes).from_address(dst)
+                    host = torch.frombuffer(view, dtype=src.dtype).reshape(src.shape)
+                    stats["pinned_tensors"] += 1
+                    stats["mode"] = "pinned"
+                    return host
+        stats["mode"] = stats["mode"] if stats["mode"] == "pinned" else "torch"
+        return t.cpu()
+
+    def walk(obj):
+        if isinstance(obj, torch.Tensor):
+            return to_host(obj) if obj.is_cuda else obj
+        if isinstance(obj, dict):
+            return {k: walk(v) for k, v in obj.items()}
+        if isinstance(obj, tuple):
+            vals = [walk(v) for v in obj]
+            return type(obj)(*vals) if hasattr(obj, "_fields") else tuple(vals)
+        if isinstance(obj, list):
+            return [walk(v) for v in obj]
+        return obj
+
+    return walk(result)
+
+
+def _read_frame(fd):
+    header = b""
+    while len(header) < 4:
+        chunk = os.read(fd, 4 - len(header))
+        if not chunk:
+            return None
+        header += chunk
+    (length,) = struct.unpack(">I", header)
+    if length == 0:
+        return b""
+    parts = []
+    remaining = length
+    while remaining:
+        chunk = os.read(fd, min(remaining, 1 << 20))
+        if not chunk:
+            return None
+        parts.append(chunk)
+        remaining -= len(chunk)
+    return b"".join(parts)
+
+
+def _write_frame(fd, payload):
+    os.write(fd, struct.pack(">I", len(payload)))
+    view = memoryview(payload)
+    while view:
+        written = os.write(fd, view[: 1 << 20])
+        view = view[written:]
+
+
+def _serve_one(request):
+    t0 = time.monotonic()
+    meta = {"phases_ms": {}, "gpu": _gpu_info, "staging": None,
+            "gpu_slot": GPU_SLOT,
+            "hip_visible_devices": os.environ.get("HIP_VISIBLE_DEVICES"),
+            "pid": os.getpid(), "worker": True}
+    result = None
+    exception = None
+    try:
+        fn, args, kwargs = pickle.loads(request["function_blob"])
+    except Exception as e:  # noqa: BLE001
+        exception = e
+        fn = None
+
+    home = os.getcwd()
+    if exception is None:
+        workdir = request.get("workdir") or "."
+        try:
+            os.makedirs(workdir, exist_ok=True)
+            os.chdir(workdir)
+        except OSError as e:
+            exception = e
+
+    if exception is None:
+        t_fn = time.monotonic()
+        try:
+            result = fn(*args, **kwargs)
+        except Exception as e:  # noqa: BLE001
+            exception = e
+        meta["phases_ms"]["user_fn"] = round((time.monotonic() - t_fn) * 1000, 3)
+    os.chdir(home)
+
+    if exception is None and result is not None:
+        t_stage = time.monotonic()
+        stats = {"tensors": 0, "pinned_tensors": 0, "bytes": 0, "mode": "none"}
+        try:
+            result = _stage_result(result, stats)
+            meta["staging"] = stats
+        except Exception as e:  # noqa: BLE001
+            result, exception = None, e
+        meta["phases_ms"]["staging"] = round((time.monotonic() - t_stage) * 1000, 3)
+
+    try:
+        result_blob = pickle.dumps((result, exception))
+    except Exception as e:  # noqa: BLE001
+        result_blob = pickle.dumps((None, e))
+    meta["phases_ms"]["total"] = round((time.monotonic() - t0) * 1000, 3)
+    return result_blob, meta
+
+
+def main():
+    startup_meta = {"pid": os.getpid(), "gpu": None, "error": None,
+                    "gpu_slot": GPU_SLOT,
+                    "hip_visible_devices": os.environ.get("HIP_VISIBLE_DEVICES")}
+    if GPU_LIB and GPU_SLOT is not None:
+        try:
+            _prologue()
+            startup_meta["gpu"] = _gpu_info
+        except Exception as e:  # noqa: BLE001
+            startup_meta["error"] = repr(e)
+    _write_frame(_proto_fd, pickle.dumps(("READY", startup_meta)))
+    if startup_meta["error"]:
+        sys.exit(3)
+
+    while True:
+        frame = _read_frame(0)
+        if frame is None or frame == b"":
+            break
+        request = pickle.loads(frame)
+        result_blob, meta = _serve_one(request)
+        _write_frame(_proto_fd, pickle.dumps((result_blob, meta)))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,129 @@
+"""Dispatcher-side pool of persistent remote workers.
+
+One warm worker process per GPU slot per endpoint (plus a small
+round-robin set of CPU workers when no GPU policy is active), launched
+over a long-lived transport channel and reused across electrons.  This
+removes the per-electron python + HIP-runtime start the classic stub
+pays (the reference architecture pays it per task by design —
+/root/reference/covalent_ssh_plugin/ssh.py:377-383).
+
+Module-level, keyed per endpoint: all executor instances targeting the
+same host share workers, mirroring the transport pool and slot tables.
+"""
+
+from __future__ import annotations
+
+import asyncio
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import cloudpickle
+
+from ..compat import app_log
+from ..transport.channel import Channel, ChannelClosed
+
+WorkerKey = Tuple[object, ...]
+
+
+@dataclass
+class WorkerHandle:
+    channel: Channel
+    startup_meta: dict
+    key: WorkerKey
+
+    @property
+    def alive(self) -> bool:
+        return self.channel.alive
+
+
+class WorkerStartupError(RuntimeError):
+    pass
+
+
+_workers: Dict[WorkerKey, WorkerHandle] = {}
+_locks: Dict[WorkerKey, Tuple[asyncio.Lock, object]] = {}
+_cpu_rr = 0  # round-robin cursor for CPU worker sets
+
+
+def _lock_for(key: WorkerKey) -> asyncio.Lock:
+    loop = asyncio.get_running_loop()
+    entry = _locks.get(key)
+    if entry is None or entry[1] is not loop:
+        entry = (asyncio.Lock(), loop)
+        _locks[key] = entry
+    return entry[0]
+
+
+def cpu_worker_index(pool_size: int) -> int:
+    global _cpu_rr
+    idx = _cpu_rr % max(1, pool_size)
+    _cpu_rr += 1
+    return idx
+
+
+async def get_worker(key: WorkerKey, launcher, startup_timeout: float = 180.0) -> WorkerHandle:
+    """Return the live worker for ``key``, starting it with ``launcher``
+    (an async callable returning a Channel) if needed."""
+    loop = asyncio.get_running_loop()
+    async with _lock_for(key):
+        handle = _workers.get(key)
+        if handle is not None and handle.alive and handle.channel.loop is loop:
+            return handle
+        if handle is not None and handle.channel.loop is not loop:
+            # worker belongs to a closed event loop: reap it synchronously
+            handle.channel.kill()
+            _workers.pop(key, None)
+        channel = await launcher()
+        try:
+            ready = await channel.recv_frame(timeout=startup_timeout)
+        except (ChannelClosed, asyncio.TimeoutError) as e:
+            await channel.close()
+            raise WorkerStartupError(f"worker {key} failed to start: {e}") from e
+        tag, meta = cloudpickle.loads(ready)
+        if tag != "READY" or meta.get("error"):
+            await channel.close()
+            raise WorkerStartupError(
+                f"worker {key} startup failed: {meta.get('error')}"
+            )
+        handle = WorkerHandle(channel=channel, startup_meta=meta, key=key)
+        _workers[key] = handle
+        return handle
+
+
+async def run_task(
+    handle: WorkerHandle,
+    op_id: str,
+    workdir: str,
+    function_blob: bytes,
+    timeout: Optional[float] = None,
+):
+    """One electron through a worker.  Returns (result, exception, meta)."""
+    request = cloudpickle.dumps(
+        {"op_id": op_id, "workdir": workdir, "function_blob": function_blob}
+    )
+    reply = await handle.channel.request(request, timeout=timeout)
+    result_blob, meta = cloudpickle.loads(reply)
+    result, exception = cloudpickle.loads(result_blob)
+    return result, exception, meta
+
+
+def drop(key: WorkerKey) -> None:
+    _workers.pop(key, None)
+
+
+async def close_all() -> None:
+    handles = list(_workers.values())
+    _workers.clear()
+    _locks.clear()
+    for h in handles:
+        try:
+            await h.channel.close()
+        except Exception:  # noqa: BLE001
+            app_log.debug("worker close failed", exc_info=True)
+
+
+def reset() -> None:
+    """Synchronous test hook (leaks processes if any are live — tests
+    that start workers must close them)."""
+    _workers.clear()
+    _locks.clear()

@@ -41,7 +41,8 @@ import cloudpickle
 
 from .compat import RemoteExecutor, app_log, get_config, update_config_defaults
 from .gpu import slots as gpu_slots
-from .remote.stub import DEFAULT_STAGING_THRESHOLD, render_stub
+from .remote import workers as worker_pool
+from .remote.stub import DEFAULT_STAGING_THRESHOLD, render_stub, render_worker
 from .transport import (
     CompletedCommand,
     LocalTransport,
@@ -80,6 +81,8 @@ _EXECUTOR_PLUGIN_DEFAULTS = {
     "warmup_gpu": True,  # run the CDNA4 warm-up/probe kernel pre-task
     "pinned_staging_threshold_bytes": DEFAULT_STAGING_THRESHOLD,
     "batch_roundtrips": True,  # fused single-round-trip dispatch
+    "persistent_workers": False,  # warm worker process per GPU slot
+    "cpu_workers": 4,  # worker-set size when no GPU policy is active
 }
 
 update_config_defaults("executors.ssh", _EXECUTOR_PLUGIN_DEFAULTS)
@@ -133,6 +136,8 @@ class SSHExecutor(RemoteExecutor):
         warmup_gpu: Optional[bool] = None,
         pinned_staging_threshold_bytes: Optional[int] = None,
         batch_roundtrips: Optional[bool] = None,
+        persistent_workers: Optional[bool] = None,
+        cpu_workers: Optional[int] = None,
         fixed_gpu: int = 0,
         local_home: str = "",
     ) -> None:
@@ -173,6 +178,10 @@ class SSHExecutor(RemoteExecutor):
             _conf("pinned_staging_threshold_bytes", pinned_staging_threshold_bytes)
         )
         self.batch_roundtrips = bool(_conf("batch_roundtrips", batch_roundtrips, default=True))
+        self.persistent_workers = bool(
+            _conf("persistent_workers", persistent_workers, default=False)
+        )
+        self.cpu_workers = int(_conf("cpu_workers", cpu_workers))
         self.fixed_gpu = int(fixed_gpu)
         self.local_home = local_home
 
@@ -442,6 +451,81 @@ class SSHExecutor(RemoteExecutor):
             await transport.put_files([(local_lib, remote_lib)])
         return remote_lib
 
+    async def _provision_worker_script(self, transport: Transport, gpu_lib: str) -> str:
+        """Ship the rendered persistent-worker script once per endpoint,
+        content-addressed (re-ships automatically when config that
+        affects the rendering changes)."""
+        import hashlib
+
+        text = render_worker(
+            gpu_lib_path=gpu_lib,
+            warmup=self.warmup_gpu,
+            staging_threshold=self.pinned_staging_threshold_bytes,
+        )
+        digest = hashlib.sha256(text.encode()).hexdigest()[:12]
+        key = self._pool_key()
+        check_name = f"worker_script:{digest}"
+        cached = transport_pool.cached_check(key, check_name)
+        if cached:
+            return cached
+        remote_path = f"{self.remote_cache}/lib/worker-{digest}.py"
+        probe = await transport.run(f"test -f {shlex.quote(remote_path)}")
+        if not probe.ok:
+            Path(self.cache_dir).mkdir(parents=True, exist_ok=True)
+            # unique temp name: concurrent electrons may provision the
+            # same digest at once (idempotent, content-addressed)
+            local_tmp = os.path.join(
+                self.cache_dir, f"worker-{digest}-{uuid.uuid4().hex[:8]}.py"
+            )
+            Path(local_tmp).write_text(text)
+            try:
+                await transport.put_files([(local_tmp, remote_path)])
+            finally:
+                try:
+                    os.remove(local_tmp)
+                except OSError:
+                    pass
+        transport_pool.store_check(key, check_name, remote_path)
+        return remote_path
+
+    async def _dispatch_worker(
+        self,
+        transport: Transport,
+        operation_id: str,
+        function_blob: bytes,
+        workdir: str,
+        env: Optional[dict],
+        gpu_lib: str,
+        worker_tag: object,
+    ):
+        """Run one electron on the persistent worker for ``worker_tag``
+        (a GPU slot id, or a CPU worker index).  Respawns a dead worker
+        once before giving up."""
+        from .transport.channel import ChannelClosed
+
+        script_remote = await self._provision_worker_script(transport, gpu_lib)
+        key = (self._pool_key(), worker_tag)
+        cmd = self._wrap_conda(
+            f"{self.python_path} {shlex.quote(script_remote)}"
+        )
+
+        async def launcher():
+            return await transport.open_channel(cmd, env=env)
+
+        for attempt in (0, 1):
+            handle = await worker_pool.get_worker(key, launcher)
+            try:
+                return await worker_pool.run_task(
+                    handle, operation_id, workdir, function_blob
+                )
+            except ChannelClosed:
+                worker_pool.drop(key)
+                if attempt == 1:
+                    raise
+                app_log.warning(
+                    "worker %s died mid-task; respawning once", key
+                )
+
     # ------------------------------------------------------------------
     # Fused single-round-trip dispatch
     # ------------------------------------------------------------------
@@ -571,23 +655,45 @@ class SSHExecutor(RemoteExecutor):
                 env = slot.env()
                 record.gpu_id = slot.gpu_id
             elif policy == "fixed":
-                env = {
-                    "HIP_VISIBLE_DEVICES": str(self.fixed_gpu),
-                    "ROCR_VISIBLE_DEVICES": str(self.fixed_gpu),
-                }
+                env = {"CSP_GPU_SLOT": str(self.fixed_gpu)}
                 record.gpu_id = self.fixed_gpu
             else:
                 env = None
 
-            with timer.phase("stage"):
-                paths = self._write_function_files(
-                    operation_id, function, args, kwargs, current_remote_workdir, gpu_lib
-                )
-
             result: Any = None
             exception: Optional[Exception] = None
+            paths: Optional[Dict[str, str]] = None
 
-            if self.batch_roundtrips:
+            if self.persistent_workers:
+                with timer.phase("stage"):
+                    function_blob = cloudpickle.dumps((function, args, kwargs))
+                try:
+                    with timer.phase("dispatch"):
+                        result, exception, meta = await self._dispatch_worker(
+                            transport,
+                            operation_id,
+                            function_blob,
+                            current_remote_workdir,
+                            env,
+                            gpu_lib,
+                            (
+                                record.gpu_id
+                                if record.gpu_id is not None
+                                else f"cpu{worker_pool.cpu_worker_index(self.cpu_workers)}"
+                            ),
+                        )
+                    record.remote_meta = meta
+                except (worker_pool.WorkerStartupError, ConnectionError) as e:
+                    return await self._on_ssh_fail(
+                        function, args, kwargs, f"persistent worker failed: {e}"
+                    )
+            if not self.persistent_workers:
+                with timer.phase("stage"):
+                    paths = self._write_function_files(
+                        operation_id, function, args, kwargs, current_remote_workdir, gpu_lib
+                    )
+
+            if not self.persistent_workers and self.batch_roundtrips:
                 with timer.phase("dispatch"):
                     proc, result_bytes, meta_bytes = await self._dispatch_fused(
                         transport, paths, env
@@ -605,7 +711,7 @@ class SSHExecutor(RemoteExecutor):
                         record.load_meta(meta_bytes)
                     if self.do_cleanup is False:
                         Path(paths["result_local"]).write_bytes(result_bytes)
-            else:
+            elif not self.persistent_workers:
                 # Template path: discrete upload/submit/poll/fetch/cleanup
                 # round trips (reference §3.1 flow).
                 with timer.phase("upload"):
@@ -661,7 +767,7 @@ class SSHExecutor(RemoteExecutor):
             if slot is not None:
                 await slot.release()
 
-        if self.do_cleanup:
+        if self.do_cleanup and paths is not None:
             self._cleanup_local(paths)
 
         record.phases = timer.snapshot()
@@ -687,4 +793,5 @@ class SSHExecutor(RemoteExecutor):
     # outlive tasks by design.
     @staticmethod
     async def close_pool() -> None:
+        await worker_pool.close_all()
         await transport_pool.close_all()

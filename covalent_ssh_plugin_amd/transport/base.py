@@ -107,6 +107,12 @@ class Transport(ABC):
     @abstractmethod
     def is_connected(self) -> bool: ...
 
+    @abstractmethod
+    async def open_channel(self, command: str, env: Optional[dict] = None):
+        """Start a long-lived remote process and return a
+        :class:`~covalent_ssh_plugin_amd.transport.channel.Channel` to
+        its stdin/stdout (used for persistent workers)."""
+
     # -- helpers shared by implementations ---------------------------------
 
     @staticmethod

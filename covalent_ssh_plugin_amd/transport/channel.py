@@ -1,0 +1,99 @@
+"""Long-lived framed channels over a transport.
+
+A :class:`Channel` wraps one persistent remote process (a worker,
+remote/worker_template.py) whose stdin/stdout carry 4-byte
+length-prefixed frames.  Channels are how persistent workers are fed:
+over SSH the process rides the pooled ControlMaster (no per-task
+handshake), locally it is a plain subprocess — either way the remote
+python + HIP runtime stay warm across electrons.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import struct
+from typing import List, Optional
+
+
+class ChannelClosed(ConnectionError):
+    """The remote worker process went away (EOF on its stdout)."""
+
+
+class Channel:
+    def __init__(self, proc: asyncio.subprocess.Process, label: str = "worker"):
+        self._proc = proc
+        self._label = label
+        self._lock = asyncio.Lock()  # one in-flight request per worker
+        self.loop = asyncio.get_event_loop()
+
+    @property
+    def alive(self) -> bool:
+        return self._proc.returncode is None
+
+    async def send_frame(self, payload: bytes) -> None:
+        if not self.alive:
+            raise ChannelClosed(f"{self._label}: process exited")
+        self._proc.stdin.write(struct.pack(">I", len(payload)))
+        self._proc.stdin.write(payload)
+        await self._proc.stdin.drain()
+
+    async def recv_frame(self, timeout: Optional[float] = None) -> bytes:
+        async def _read() -> bytes:
+            header = await self._proc.stdout.readexactly(4)
+            (length,) = struct.unpack(">I", header)
+            if length == 0:
+                return b""
+            return await self._proc.stdout.readexactly(length)
+
+        try:
+            if timeout is None:
+                return await _read()
+            return await asyncio.wait_for(_read(), timeout=timeout)
+        except (asyncio.IncompleteReadError, ConnectionResetError) as e:
+            raise ChannelClosed(f"{self._label}: EOF mid-frame") from e
+
+    async def request(self, payload: bytes, timeout: Optional[float] = None) -> bytes:
+        """Serialized request/response round trip."""
+        async with self._lock:
+            await self.send_frame(payload)
+            return await self.recv_frame(timeout=timeout)
+
+    def kill(self) -> None:
+        """Synchronous hard-kill (for reaping workers whose event loop is
+        already gone)."""
+        if self.alive:
+            try:
+                self._proc.kill()
+            except ProcessLookupError:
+                pass
+
+    async def close(self) -> None:
+        if self.loop is not asyncio.get_event_loop() or self.loop.is_closed():
+            # channel belongs to another (likely closed) loop: its pipe
+            # transports cannot be driven from here — hard-kill instead
+            self.kill()
+            return
+        if self.alive:
+            try:
+                # zero-length frame = orderly shutdown
+                self._proc.stdin.write(struct.pack(">I", 0))
+                await self._proc.stdin.drain()
+                self._proc.stdin.close()
+            except (ConnectionResetError, BrokenPipeError, RuntimeError):
+                pass
+            try:
+                await asyncio.wait_for(self._proc.wait(), timeout=5)
+            except asyncio.TimeoutError:
+                self._proc.kill()
+                await self._proc.wait()
+
+
+async def open_subprocess_channel(argv: List[str], label: str) -> Channel:
+    proc = await asyncio.create_subprocess_exec(
+        *argv,
+        stdin=asyncio.subprocess.PIPE,
+        stdout=asyncio.subprocess.PIPE,
+        stderr=None,  # worker/user stderr passes through to our logs
+        limit=64 * 1024 * 1024,
+    )
+    return Channel(proc, label=label)

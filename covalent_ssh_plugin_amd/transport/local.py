@@ -89,5 +89,21 @@ class LocalTransport(Transport):
         Path(local_path).parent.mkdir(parents=True, exist_ok=True)
         await asyncio.to_thread(shutil.copyfile, src, local_path)
 
+    async def open_channel(self, command: str, env: Optional[dict] = None):
+        from .channel import Channel
+
+        full_cmd = self._env_prefix(env) + command
+        proc = await asyncio.create_subprocess_exec(
+            "bash",
+            "-c",
+            full_cmd,
+            cwd=str(self._home),
+            stdin=asyncio.subprocess.PIPE,
+            stdout=asyncio.subprocess.PIPE,
+            stderr=None,
+            limit=64 * 1024 * 1024,
+        )
+        return Channel(proc, label=f"local-worker")
+
     async def close(self) -> None:
         self._connected = False

@@ -174,6 +174,19 @@ class OpenSSHTransport(Transport):
         Path(local_path).parent.mkdir(parents=True, exist_ok=True)
         Path(local_path).write_bytes(result.stdout)
 
+    async def open_channel(self, command: str, env: Optional[dict] = None):
+        from .channel import Channel
+
+        argv = self._base_args() + [self.hostname, "--", self._env_prefix(env) + command]
+        proc = await asyncio.create_subprocess_exec(
+            *argv,
+            stdin=asyncio.subprocess.PIPE,
+            stdout=asyncio.subprocess.PIPE,
+            stderr=None,
+            limit=64 * 1024 * 1024,
+        )
+        return Channel(proc, label=f"worker@{self.endpoint}")
+
     async def close(self) -> None:
         if not self._connected:
             return

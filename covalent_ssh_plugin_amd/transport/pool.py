@@ -24,15 +24,20 @@ PoolKey = Tuple[str, ...]
 
 _pool: Dict[PoolKey, Transport] = {}
 _pool_lock: Optional[asyncio.Lock] = None
+_pool_lock_loop = None
 # One-time environment check results, keyed by (pool_key, check_name).
 _env_checks: Dict[Tuple[PoolKey, str], object] = {}
 
 
 def _lock() -> asyncio.Lock:
-    # Created lazily so the lock binds to the running loop.
-    global _pool_lock
-    if _pool_lock is None:
+    # Created lazily so the lock binds to the RUNNING loop; recreated if
+    # a new asyncio.run() started a fresh loop (the old loop is closed,
+    # so it can have no live waiters).
+    global _pool_lock, _pool_lock_loop
+    loop = asyncio.get_running_loop()
+    if _pool_lock is None or _pool_lock_loop is not loop:
         _pool_lock = asyncio.Lock()
+        _pool_lock_loop = loop
     return _pool_lock
 
 

@@ -21,13 +21,16 @@ def _fresh_pools():
     """Isolate module-level pooled state (transports, GPU slot tables)
     between tests."""
     from covalent_ssh_plugin_amd.gpu import slots
+    from covalent_ssh_plugin_amd.remote import workers
     from covalent_ssh_plugin_amd.transport import pool
 
     pool.reset()
     slots.reset()
+    workers.reset()
     yield
-    # Close any transports a test left connected.
+    # Close any workers/transports a test left running.
     try:
+        asyncio.run(workers.close_all())
         asyncio.run(pool.close_all())
     except RuntimeError:
         pass

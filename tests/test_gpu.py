@@ -69,21 +69,32 @@ def test_pinned_staging_correctness(gpu_lib):
 
 def test_pinned_staging_bandwidth(gpu_lib):
     """Pinned D2H must beat a conservative pageable floor (PCIe Gen5 x16
-    pinned should sustain tens of GB/s)."""
+    pinned should sustain tens of GB/s).  The pool is warmed at full size
+    first so the one-time hipHostMalloc pinning cost is excluded."""
+    import ctypes
     import time
-
-    from covalent_ssh_plugin_amd.gpu import probe
 
     nbytes = 1 << 30  # 1 GiB
     t = torch.empty(nbytes, dtype=torch.uint8, device="cuda")
     torch.cuda.synchronize()
-    probe.staged_d2h_bytes(t.data_ptr(), 1 << 20)  # warm the pool
+    dst = gpu_lib.csp_staging_get(nbytes)  # warm: pin the full buffer once
+    assert dst
+    gpu_lib.csp_memcpy_d2h(ctypes.c_void_p(dst), ctypes.c_void_p(t.data_ptr()), nbytes)
     t0 = time.perf_counter()
-    probe.staged_d2h_bytes(t.data_ptr(), nbytes)
+    rc = gpu_lib.csp_memcpy_d2h(
+        ctypes.c_void_p(dst), ctypes.c_void_p(t.data_ptr()), nbytes
+    )
     dt = time.perf_counter() - t0
+    assert rc == 0
     gbps = nbytes / dt / 1e9
     print(f"pinned D2H: {gbps:.1f} GB/s")
-    assert gbps > 10.0, gbps
+    assert gbps > 20.0, gbps
+
+    # pageable comparison (torch .cpu() into fresh pageable memory)
+    t0 = time.perf_counter()
+    t.cpu()
+    dt_pageable = time.perf_counter() - t0
+    print(f"pageable D2H: {nbytes / dt_pageable / 1e9:.1f} GB/s")
 
 
 def test_host_alloc_roundtrip(gpu_lib):
@@ -132,7 +143,7 @@ def test_executor_gpu_electron_with_prologue(tmp_path):
     assert meta is not None
     assert meta["gpu"] is not None and "gfx950" in meta["gpu"]["gcn_arch"]
     assert meta["staging"]["mode"] == "pinned", meta
-    assert meta["hip_visible_devices"] == str(rec.gpu_id)
+    assert meta["gpu_slot"] == str(rec.gpu_id)
     asyncio.run(ex.close_pool())
 
 
@@ -169,7 +180,7 @@ def test_concurrent_electrons_share_gpu_slots(tmp_path):
     def whoami():
         import os
 
-        return os.environ.get("HIP_VISIBLE_DEVICES")
+        return os.environ.get("CSP_GPU_SLOT")
 
     ex = _executor(tmp_path, warmup_gpu=False)
 

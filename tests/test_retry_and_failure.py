@@ -39,6 +39,9 @@ class FlakyTransport(Transport):
     async def get_file(self, remote_path, local_path):
         pass
 
+    async def open_channel(self, command, env=None):
+        raise NotImplementedError
+
     async def close(self):
         self._connected = False
 

@@ -74,8 +74,7 @@ def test_env_injection():
         table = SlotTable(num_gpus=8)
         s = await table.acquire()
         env = s.env()
-        assert env["HIP_VISIBLE_DEVICES"] == str(s.gpu_id)
-        assert env["ROCR_VISIBLE_DEVICES"] == str(s.gpu_id)
+        assert env["CSP_GPU_SLOT"] == str(s.gpu_id)
         await s.release()
 
     asyncio.run(main())

@@ -115,7 +115,7 @@ def test_missing_gpu_lib_with_slot_is_loud(tmp_path):
     )
     script_file.write_text(script)
     env = dict(os.environ)
-    env["HIP_VISIBLE_DEVICES"] = "0"
+    env["CSP_GPU_SLOT"] = "0"
     proc = subprocess.run(
         [sys.executable, str(script_file)], capture_output=True, cwd=tmp_path, env=env, timeout=60
     )
@@ -123,3 +123,35 @@ def test_missing_gpu_lib_with_slot_is_loud(tmp_path):
     result, exception = pickle.loads(result_file.read_bytes())
     assert result is None
     assert exception is not None  # OSError from ctypes.CDLL
+
+
+def test_gpu_slot_resolves_within_ambient_visibility(tmp_path):
+    """CSP_GPU_SLOT must select WITHIN an ambient HIP_VISIBLE_DEVICES
+    list (pod GPU isolation), not clobber it."""
+    import os
+
+    def visible():
+        import os
+
+        return os.environ.get("HIP_VISIBLE_DEVICES")
+
+    env = dict(os.environ)
+    env["HIP_VISIBLE_DEVICES"] = "5,7"
+    env["CSP_GPU_SLOT"] = "1"
+    proc, result_file, _ = _run_stub(tmp_path, visible, env=env)
+    assert proc.returncode == 0
+    result, exception = pickle.loads(result_file.read_bytes())
+    assert exception is None
+    assert result == "7"
+
+    env["CSP_GPU_SLOT"] = "0"
+    proc, result_file, _ = _run_stub(tmp_path, visible, env=env)
+    result, _ = pickle.loads(result_file.read_bytes())
+    assert result == "5"
+
+    # no ambient restriction -> slot index used directly
+    env.pop("HIP_VISIBLE_DEVICES")
+    env["CSP_GPU_SLOT"] = "3"
+    proc, result_file, _ = _run_stub(tmp_path, visible, env=env)
+    result, _ = pickle.loads(result_file.read_bytes())
+    assert result == "3"

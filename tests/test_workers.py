@@ -1,0 +1,116 @@
+"""Persistent worker mode: warm worker processes fed over framed
+channels — the executor's high-throughput dispatch path."""
+
+import asyncio
+
+import pytest
+
+
+def test_worker_roundtrip_and_reuse(local_executor):
+    ex = local_executor(persistent_workers=True, cpu_workers=1)
+
+    def fn(x):
+        import os
+
+        return (x * 2, os.getpid())
+
+    async def main():
+        r1 = await ex.execute(fn, [21], {}, dispatch_id="w", node_id=0)
+        r2 = await ex.execute(fn, [100], {}, dispatch_id="w", node_id=1)
+        return r1, r2
+
+    (r1, pid1), (r2, pid2) = asyncio.run(main())
+    assert (r1, r2) == (42, 200)
+    # same worker process served both electrons (warm reuse)
+    assert pid1 == pid2
+    assert ex.last_task_record.remote_meta["worker"] is True
+
+
+def test_worker_exception_roundtrip(local_executor):
+    ex = local_executor(persistent_workers=True)
+
+    def boom():
+        raise ValueError("worker task failed")
+
+    with pytest.raises(ValueError, match="worker task failed"):
+        asyncio.run(ex.execute(boom, [], {}))
+
+    # worker survives a task exception and serves the next electron
+    def ok():
+        return "fine"
+
+    assert asyncio.run(ex.execute(ok, [], {})) == "fine"
+
+
+def test_worker_stdout_pollution_safe(local_executor):
+    ex = local_executor(persistent_workers=True)
+
+    def chatty():
+        print("x" * 10000)
+        print("\x00\xff frames? \x00\x00\x00\x05")
+        return 7
+
+    assert asyncio.run(ex.execute(chatty, [], {})) == 7
+
+
+def test_worker_concurrency_cpu_pool(local_executor):
+    ex = local_executor(persistent_workers=True, cpu_workers=4)
+
+    def slow(i):
+        import time
+
+        time.sleep(0.2)
+        return i
+
+    async def main():
+        import time
+
+        t0 = time.perf_counter()
+        out = await asyncio.gather(
+            *[ex.execute(slow, [i], {}, dispatch_id="c", node_id=i) for i in range(4)]
+        )
+        return out, time.perf_counter() - t0
+
+    out, dt = asyncio.run(main())
+    assert out == [0, 1, 2, 3]
+    # 4 workers -> 4x 0.2s sleeps overlap; generous bound for slow CI
+    assert dt < 0.75, dt
+
+
+def test_worker_death_respawn(local_executor):
+    ex = local_executor(persistent_workers=True, cpu_workers=1)
+
+    def suicide():
+        import os
+
+        os._exit(9)  # hard-kill the worker mid-task
+
+    def fine():
+        return 5
+
+    with pytest.raises(RuntimeError):
+        asyncio.run(ex.execute(suicide, [], {}, dispatch_id="k", node_id=0))
+    # a fresh worker replaces the dead one
+    assert asyncio.run(ex.execute(fine, [], {}, dispatch_id="k", node_id=1)) == 5
+
+
+def test_worker_workdir(local_executor):
+    ex = local_executor(persistent_workers=True, create_unique_workdir=True)
+
+    def cwd():
+        import os
+
+        return os.getcwd()
+
+    out = asyncio.run(ex.execute(cwd, [], {}, dispatch_id="wd", node_id=3))
+    assert out.endswith("covalent-workdir/wd/node_3")
+
+
+def test_worker_large_payload(local_executor):
+    ex = local_executor(persistent_workers=True)
+
+    def big(n):
+        return b"z" * n
+
+    out = asyncio.run(ex.execute(big, [8 * 1024 * 1024], {}))
+    assert len(out) == 8 * 1024 * 1024
