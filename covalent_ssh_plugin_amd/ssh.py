@@ -788,6 +788,13 @@ class SSHExecutor(RemoteExecutor):
             except OSError:
                 pass
 
+    def stats(self) -> dict:
+        """Latency percentiles + per-phase means over this executor's
+        completed tasks (observability; SURVEY.md §5 metrics row)."""
+        from .utils.timing import summarize
+
+        return summarize(self.task_records)
+
     # Convenience for closing pooled transports (e.g. at interpreter exit
     # or between tests).  The reference closes per-task; pooled transports
     # outlive tasks by design.

@@ -57,3 +57,32 @@ class TaskRecord:
 
     def load_meta_file(self, path: str) -> None:
         self.load_meta(Path(path).read_bytes())
+
+
+def summarize(records) -> dict:
+    """Aggregate a list of TaskRecords into dispatch statistics
+    (throughput over the records' total span is the caller's job — this
+    reports latency percentiles and phase means)."""
+    import statistics
+
+    if not records:
+        return {"count": 0}
+    totals = [r.total_s for r in records]
+    phases: Dict[str, list] = {}
+    for r in records:
+        for name, v in r.phases.items():
+            phases.setdefault(name, []).append(v)
+    out = {
+        "count": len(records),
+        "p50_ms": statistics.median(totals) * 1000.0,
+        "p90_ms": sorted(totals)[max(0, int(len(totals) * 0.9) - 1)] * 1000.0,
+        "mean_ms": statistics.fmean(totals) * 1000.0,
+        "phase_mean_ms": {
+            k: statistics.fmean(v) * 1000.0 for k, v in phases.items()
+        },
+        "gpu_spread": {},
+    }
+    for r in records:
+        if r.gpu_id is not None:
+            out["gpu_spread"][r.gpu_id] = out["gpu_spread"].get(r.gpu_id, 0) + 1
+    return out

@@ -174,3 +174,17 @@ def test_concurrent_electrons(local_executor):
     results = asyncio.run(main())
     assert results == [2 * i for i in range(12)]
     assert len(ex.task_records) == 12
+
+
+def test_stats_summary(local_executor):
+    ex = local_executor()
+
+    async def main():
+        for i in range(4):
+            await ex.execute(_add, [i, 1], {}, dispatch_id="st", node_id=i)
+
+    asyncio.run(main())
+    s = ex.stats()
+    assert s["count"] == 4
+    assert s["p50_ms"] > 0
+    assert "dispatch" in s["phase_mean_ms"]
