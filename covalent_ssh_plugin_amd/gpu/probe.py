@@ -46,6 +46,17 @@ def load(path: str = "") -> ctypes.CDLL:
             f"{_LIB_NAME} not built — run __graft_entry__.build() / "
             "python -m covalent_ssh_plugin_amd.ops.build"
         )
+    # One HSA runtime per process: bind torch's bundled ROCm runtime
+    # first when torch is installed, so this library shares it instead of
+    # initializing the system runtime (which would break a later
+    # torch.cuda init with "No HIP GPUs are available").
+    import importlib.util
+
+    if importlib.util.find_spec("torch") is not None:
+        import torch
+
+        if torch.cuda.is_available():
+            torch.cuda.init()
     lib = ctypes.CDLL(lib_path)
 
     lib.csp_device_count.restype = ctypes.c_int

@@ -83,15 +83,54 @@ __global__ __launch_bounds__(256) void csp_mfma_spin_kernel(
 
 // HBM sweep: grid-stride float4 copy (read 16 B + write 16 B per lane per
 // element).  Doubles as the bandwidth probe and the HBM warm-up.
-__global__ __launch_bounds__(256) void csp_hbm_sweep_kernel(
+// Template knobs measured on-box (tools: csp_hbm_bench):
+//   UNROLL — independent loads in flight per lane (ILP on top of the
+//            grid's TLP; HBM latency ~900 cyc wants >1 per lane),
+//   NT     — nontemporal load+store (aux=2): a pure streaming copy never
+//            re-reads, so bypassing L2 residency avoids evicting it.
+template <int UNROLL, bool NT>
+__global__ __launch_bounds__(256) void csp_hbm_sweep_t(
     const float4* __restrict__ src, float4* __restrict__ dst, size_t n4) {
+    const size_t stride = (size_t)gridDim.x * blockDim.x;
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
-    size_t stride = (size_t)gridDim.x * blockDim.x;
-    for (; i < n4; i += stride) {
-        float4 v = src[i];
-        dst[i] = v;
+    // nontemporal builtins need a true vector type, not HIP's float4 class
+    typedef float f32x4v __attribute__((ext_vector_type(4)));
+    const f32x4v* __restrict__ srcv = reinterpret_cast<const f32x4v*>(src);
+    f32x4v* __restrict__ dstv = reinterpret_cast<f32x4v*>(dst);
+    for (; i + (UNROLL - 1) * stride < n4; i += UNROLL * stride) {
+        f32x4v v[UNROLL];
+#pragma unroll
+        for (int u = 0; u < UNROLL; ++u)
+            v[u] = NT ? __builtin_nontemporal_load(&srcv[i + u * stride])
+                      : srcv[i + u * stride];
+#pragma unroll
+        for (int u = 0; u < UNROLL; ++u) {
+            if (NT)
+                __builtin_nontemporal_store(v[u], &dstv[i + u * stride]);
+            else
+                dstv[i + u * stride] = v[u];
+        }
+    }
+    for (; i < n4; i += stride) dst[i] = src[i];
+}
+
+using sweep_fn = void (*)(const float4*, float4*, size_t);
+
+static sweep_fn sweep_variant(int variant) {
+    switch (variant) {
+        case 1: return csp_hbm_sweep_t<4, false>;
+        case 2: return csp_hbm_sweep_t<1, true>;
+        case 3: return csp_hbm_sweep_t<4, true>;
+        case 4: return csp_hbm_sweep_t<8, true>;
+        default: return csp_hbm_sweep_t<1, false>;
     }
 }
+
+// Default variant for probe/warm-up; selected from on-box measurements
+// (profiles/hbm_sweep_variants.md).
+static int g_sweep_variant = 3;
+
+extern "C" void csp_set_sweep_variant(int v) { g_sweep_variant = v; }
 
 // ---------------------------------------------------------------------------
 // Device / measurement helpers
@@ -123,7 +162,8 @@ static int run_mfma_spin(int iters_per_wave, int blocks, float* ms_out) {
 }
 
 static int run_hbm_sweep(float4* buf_a, float4* buf_b, size_t n4, int reps,
-                         float* ms_out) {
+                         float* ms_out, int variant = -1) {
+    sweep_fn kern = sweep_variant(variant < 0 ? g_sweep_variant : variant);
     hipEvent_t t0, t1;
     HIP_TRY(hipEventCreate(&t0));
     HIP_TRY(hipEventCreate(&t1));
@@ -132,8 +172,7 @@ static int run_hbm_sweep(float4* buf_a, float4* buf_b, size_t n4, int reps,
         // ping-pong so reads are never served from a just-written line
         const float4* s = (r & 1) ? buf_b : buf_a;
         float4* d = (r & 1) ? buf_a : buf_b;
-        hipLaunchKernelGGL(csp_hbm_sweep_kernel, dim3(4096), dim3(256), 0, 0,
-                           s, d, n4);
+        hipLaunchKernelGGL(kern, dim3(4096), dim3(256), 0, 0, s, d, n4);
     }
     HIP_TRY(hipEventRecord(t1));
     HIP_TRY(hipEventSynchronize(t1));
@@ -302,5 +341,25 @@ extern "C" int csp_memcpy_h2d(void* dst_dev, const void* src_host, size_t n) {
     hipStream_t s = copy_stream();
     HIP_TRY(hipMemcpyAsync(dst_dev, src_host, n, hipMemcpyHostToDevice, s));
     HIP_TRY(hipStreamSynchronize(s));
+    return 0;
+}
+
+// On-box variant exploration for the HBM sweep (tools only; the probe
+// uses g_sweep_variant).  Returns measured GB/s via *gbps.
+extern "C" int csp_hbm_bench(int device, int variant, size_t bytes, int reps,
+                             double* gbps) {
+    HIP_TRY(hipSetDevice(device));
+    size_t n4 = bytes / sizeof(float4);
+    float4 *a = nullptr, *b = nullptr;
+    HIP_TRY(hipMalloc(&a, bytes));
+    HIP_TRY(hipMalloc(&b, bytes));
+    HIP_TRY(hipMemsetAsync(a, 1, bytes));
+    float warm_ms = 0.f, ms = 0.f;
+    int rc = run_hbm_sweep(a, b, n4, 2, &warm_ms, variant);
+    if (rc == 0) rc = run_hbm_sweep(a, b, n4, reps, &ms, variant);
+    (void)hipFree(a);
+    (void)hipFree(b);
+    if (rc != 0) return rc;
+    *gbps = (2.0 * (double)bytes * reps / 1.0e9) / ((double)ms / 1000.0);
     return 0;
 }

@@ -97,8 +97,29 @@ except Exception as e:  # noqa: BLE001
     sys.exit(1)
 
 
+def _ensure_torch_runtime_first():
+    """One process can host only ONE HSA runtime.  torch bundles its own
+    ROCm runtime; if our ctypes library initializes the system runtime
+    first, a later torch.cuda init finds no agents ("No HIP GPUs are
+    available").  So when torch is installed, import it (binding its
+    runtime) BEFORE loading the CDNA4 library — the library then resolves
+    libamdhip64.so.7 to torch's already-loaded copy (soname dedup) and
+    both share one runtime.  Measured on-box: probe-then-torch breaks,
+    torch-then-probe runs the probe at full rate."""
+    import importlib.util
+
+    if importlib.util.find_spec("torch") is None:
+        return
+    import torch
+
+    if torch.cuda.is_available():
+        torch.cuda.init()
+
+
 def _load_gpu_lib():
     import ctypes
+
+    _ensure_torch_runtime_first()
 
     lib = ctypes.CDLL(GPU_LIB)
     lib.csp_probe_json.restype = ctypes.c_int
