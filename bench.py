@@ -39,7 +39,7 @@ if REPO_ROOT not in sys.path:
 def build_electron(config: str):
     """Electron bodies are defined inside a factory so cloudpickle
     serializes them by value."""
-    if config in ("noop", "noop-stub"):
+    if config in ("noop", "noop-stub", "fan"):
 
         def noop():
             return 0
@@ -94,6 +94,8 @@ def main() -> None:
     parser.add_argument("--steps", type=int, default=32)
     parser.add_argument("--warmup", type=int, default=8)
     parser.add_argument("--config", default="noop")
+    parser.add_argument("--fan", type=int, default=64,
+                        help="concurrent electrons per step for --config fan")
     args = parser.parse_args()
 
     import torch
@@ -132,7 +134,10 @@ def main() -> None:
                 local_home=home,
                 cache_dir=cache,
                 python_path=sys.executable,
-                hip_visible_devices_policy="fixed" if has_cuda else "none",
+                hip_visible_devices_policy=(
+                    ("roundrobin" if args.config == "fan" else "fixed")
+                    if has_cuda else "none"
+                ),
                 fixed_gpu=local_rank,
                 gpu_slots=max(1, torch.cuda.device_count()) if has_cuda else 1,
                 # dispatch-throughput metric: slot pinning yes, per-task
@@ -146,10 +151,21 @@ def main() -> None:
 
             async def one_step() -> float:
                 t0 = time.perf_counter()
-                await ex.execute(
-                    fn, list(fargs), dict(fkwargs),
-                    dispatch_id=f"bench{rank}", node_id=0,
-                )
+                if args.config == "fan":
+                    # BASELINE config 3: a fan of no-op electrons spread
+                    # round-robin across the node's GPU slots
+                    await asyncio.gather(
+                        *[
+                            ex.execute(fn, list(fargs), dict(fkwargs),
+                                       dispatch_id=f"bench{rank}", node_id=i)
+                            for i in range(args.fan)
+                        ]
+                    )
+                else:
+                    await ex.execute(
+                        fn, list(fargs), dict(fkwargs),
+                        dispatch_id=f"bench{rank}", node_id=0,
+                    )
                 return time.perf_counter() - t0
 
             for _ in range(args.warmup):
@@ -174,7 +190,8 @@ def main() -> None:
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    total_electrons = args.steps * world_size
+    electrons_per_step = args.fan if args.config == "fan" else 1
+    total_electrons = args.steps * world_size * electrons_per_step
     eps = total_electrons / elapsed
     p50_ms = statistics.median(out["latencies"]) * 1000.0
     ms_per_step = elapsed / args.steps * 1000.0
