@@ -70,6 +70,9 @@ def load(path: str = "") -> ctypes.CDLL:
     lib.csp_host_free.argtypes = [ctypes.c_void_p]
     lib.csp_staging_get.restype = ctypes.c_void_p
     lib.csp_staging_get.argtypes = [ctypes.c_size_t]
+    lib.csp_staging_alloc.restype = ctypes.c_void_p
+    lib.csp_staging_alloc.argtypes = [ctypes.c_size_t]
+    lib.csp_staging_release_all.restype = ctypes.c_int
     lib.csp_staging_reset.restype = ctypes.c_int
     lib.csp_memcpy_d2h.restype = ctypes.c_int
     lib.csp_memcpy_d2h.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_size_t]
@@ -114,12 +117,15 @@ def staged_d2h_bytes(data_ptr: int, nbytes: int) -> bytes:
     """Copy ``nbytes`` from device pointer ``data_ptr`` to host through
     the pooled pinned staging buffer and return them as bytes."""
     lib = load()
-    dst = lib.csp_staging_get(nbytes)
+    dst = lib.csp_staging_alloc(nbytes)
     if not dst:
-        _check(lib, -1, "csp_staging_get")
-    _check(
-        lib,
-        lib.csp_memcpy_d2h(ctypes.c_void_p(dst), ctypes.c_void_p(data_ptr), nbytes),
-        "csp_memcpy_d2h",
-    )
-    return ctypes.string_at(dst, nbytes)
+        _check(lib, -1, "csp_staging_alloc")
+    try:
+        _check(
+            lib,
+            lib.csp_memcpy_d2h(ctypes.c_void_p(dst), ctypes.c_void_p(data_ptr), nbytes),
+            "csp_memcpy_d2h",
+        )
+        return ctypes.string_at(dst, nbytes)
+    finally:
+        lib.csp_staging_release_all()

@@ -29,7 +29,10 @@
 
 #include <cstdio>
 #include <cstring>
+#include <algorithm>
 #include <mutex>
+#include <vector>
+#include <utility>
 
 // ---------------------------------------------------------------------------
 // Error plumbing
@@ -281,23 +284,31 @@ extern "C" int csp_host_free(void* p) {
     return 0;
 }
 
-// Grow-only pooled staging buffer: electrons in one stub process (and the
-// future persistent worker) reuse a single pinned allocation.
+// Pooled pinned staging allocator.  A task result may contain several
+// large tensors, each needing its own live pinned block until the reply
+// is written — so this is a freelist of hipHostMalloc blocks, recycled
+// across electrons (pinning 1 GiB costs ~100s of ms; reuse makes it
+// one-time).  csp_staging_release_all() returns every outstanding block
+// to the freelist; blocks beyond the cache cap are actually freed.
 static std::mutex g_staging_mu;
-static void* g_staging_buf = nullptr;
-static size_t g_staging_cap = 0;
+static std::vector<std::pair<void*, size_t>> g_staging_free;
+static std::vector<std::pair<void*, size_t>> g_staging_in_use;
+static const size_t kStagingCacheCap = size_t(6) << 30;  // 6 GiB pinned cache
 
-extern "C" void* csp_staging_get(size_t nbytes) {
+extern "C" void* csp_staging_alloc(size_t nbytes) {
     std::lock_guard<std::mutex> lock(g_staging_mu);
-    if (nbytes <= g_staging_cap) return g_staging_buf;
-    if (g_staging_buf) {
-        hipError_t e = hipHostFree(g_staging_buf);
-        g_staging_buf = nullptr;
-        g_staging_cap = 0;
-        if (e != hipSuccess) {
-            set_err("hipHostFree(staging)", e);
-            return nullptr;
-        }
+    // best-fit from the freelist
+    int best = -1;
+    for (int i = 0; i < (int)g_staging_free.size(); ++i) {
+        if (g_staging_free[i].second >= nbytes &&
+            (best < 0 || g_staging_free[i].second < g_staging_free[best].second))
+            best = i;
+    }
+    if (best >= 0) {
+        auto blk = g_staging_free[best];
+        g_staging_free.erase(g_staging_free.begin() + best);
+        g_staging_in_use.push_back(blk);
+        return blk.first;
     }
     void* p = nullptr;
     hipError_t e = hipHostMalloc(&p, nbytes, hipHostMallocDefault);
@@ -305,16 +316,40 @@ extern "C" void* csp_staging_get(size_t nbytes) {
         set_err("hipHostMalloc(staging)", e);
         return nullptr;
     }
-    g_staging_buf = p;
-    g_staging_cap = nbytes;
+    g_staging_in_use.push_back({p, nbytes});
     return p;
+}
+
+// Back-compat alias (same semantics as alloc).
+extern "C" void* csp_staging_get(size_t nbytes) { return csp_staging_alloc(nbytes); }
+
+extern "C" int csp_staging_release_all() {
+    std::lock_guard<std::mutex> lock(g_staging_mu);
+    for (auto& blk : g_staging_in_use) g_staging_free.push_back(blk);
+    g_staging_in_use.clear();
+    // trim the cache: keep the largest blocks up to the cap
+    std::sort(g_staging_free.begin(), g_staging_free.end(),
+              [](auto& a, auto& b) { return a.second > b.second; });
+    size_t kept = 0;
+    std::vector<std::pair<void*, size_t>> keep;
+    for (auto& blk : g_staging_free) {
+        if (kept + blk.second <= kStagingCacheCap) {
+            kept += blk.second;
+            keep.push_back(blk);
+        } else {
+            HIP_TRY(hipHostFree(blk.first));
+        }
+    }
+    g_staging_free.swap(keep);
+    return 0;
 }
 
 extern "C" int csp_staging_reset() {
     std::lock_guard<std::mutex> lock(g_staging_mu);
-    if (g_staging_buf) HIP_TRY(hipHostFree(g_staging_buf));
-    g_staging_buf = nullptr;
-    g_staging_cap = 0;
+    for (auto& blk : g_staging_free) HIP_TRY(hipHostFree(blk.first));
+    for (auto& blk : g_staging_in_use) HIP_TRY(hipHostFree(blk.first));
+    g_staging_free.clear();
+    g_staging_in_use.clear();
     return 0;
 }
 

@@ -17,8 +17,14 @@ python + HIP-runtime start per electron (the reference's architecture,
 
 Protocol (all frames: 4-byte big-endian length + payload):
   request  = pickle dict {"op_id", "workdir", "function_blob"}
-  response = pickle tuple (result_blob: bytes, meta: dict)
-  a zero-length request frame means: shut down.
+  response = pickle tuple ("R1", result_blob: bytes, meta: dict, nbuf: int)
+             followed by nbuf RAW tensor-buffer frames.  Large tensors in
+             the result are replaced by ("__csp_tensor_buffer_v1__", i)
+             markers and shipped out-of-band as raw frames — CUDA tensors
+             straight from the hipHostMalloc-pinned staging block, large
+             CPU tensors zero-copy from their own storage — skipping both
+             pickle copies on the payload.  meta["buffers"][i] carries
+             dtype/shape.  A zero-length request frame means: shut down.
 
 The worker's REAL stdout is reserved for the protocol; fd 1 is
 re-pointed at stderr before any user code runs, so user prints cannot
@@ -98,8 +104,9 @@ def _load_gpu_lib():
     lib.csp_probe_json.argtypes = [ctypes.c_int, ctypes.c_char_p, ctypes.c_size_t]
     lib.csp_warmup.restype = ctypes.c_int
     lib.csp_warmup.argtypes = [ctypes.c_int, ctypes.c_int]
-    lib.csp_staging_get.restype = ctypes.c_void_p
-    lib.csp_staging_get.argtypes = [ctypes.c_size_t]
+    lib.csp_staging_alloc.restype = ctypes.c_void_p
+    lib.csp_staging_alloc.argtypes = [ctypes.c_size_t]
+    lib.csp_staging_release_all.restype = ctypes.c_int
     lib.csp_memcpy_d2h.restype = ctypes.c_int
     lib.csp_memcpy_d2h.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_size_t]
     lib.csp_last_error.restype = ctypes.c_char_p
@@ -129,8 +136,11 @@ def _prologue():
             )
 
 
-def _stage_result(result, stats):
-    """Pinned-staging of CUDA tensors (same policy as the stub)."""
+def _stage_result(result, stats, buffers, buffer_meta):
+    """Replace large tensors in ``result`` with out-of-band buffer
+    markers; CUDA tensors are staged through hipHostMalloc-pinned memory
+    (fast D2H), large contiguous CPU tensors are shipped zero-copy from
+    their own storage.  Small tensors stay inline in the pickle."""
     if "torch" not in sys.modules:
         return result
     import ctypes
@@ -144,30 +154,49 @@ def _stage_result(result, stats):
         except Exception:  # noqa: BLE001
             lib = None
 
+    def emit_buffer(view, keepalive, dtype, shape):
+        buffers.append((view, keepalive))
+        buffer_meta.append(
+            {"dtype": str(dtype).replace("torch.", ""), "shape": list(shape)}
+        )
+        return ("__csp_tensor_buffer_v1__", len(buffers) - 1)
+
     def to_host(t):
         stats["tensors"] += 1
         nbytes = t.numel() * t.element_size()
         stats["bytes"] += nbytes
-        if lib is not None and nbytes >= STAGING_THRESHOLD:
-            src = t.contiguous()
-            torch.cuda.synchronize()
-            dst = lib.csp_staging_get(nbytes)
-            if dst:
-                rc = lib.csp_memcpy_d2h(
-                    ctypes.c_void_p(dst), ctypes.c_void_p(src.data_ptr()), nbytes
+        if nbytes < STAGING_THRESHOLD:
+            return t.cpu() if t.is_cuda else t
+        if t.is_cuda:
+            if lib is None:
+                raise RuntimeError(
+                    "CUDA tensor result but the CDNA4 staging library is "
+                    "not available on this GPU host"
                 )
-                if rc == 0:
-                    view = (ctypes.c_char * nbytes).from_address(dst)
-                    host = torch.frombuffer(view, dtype=src.dtype).reshape(src.shape)
-                    stats["pinned_tensors"] += 1
-                    stats["mode"] = "pinned"
-                    return host
-        stats["mode"] = stats["mode"] if stats["mode"] == "pinned" else "torch"
-        return t.cpu()
+            src_t = t.contiguous()
+            torch.cuda.synchronize()
+            dst = lib.csp_staging_alloc(nbytes)
+            if not dst:
+                raise RuntimeError("csp_staging_alloc failed")
+            rc = lib.csp_memcpy_d2h(
+                ctypes.c_void_p(dst), ctypes.c_void_p(src_t.data_ptr()), nbytes
+            )
+            if rc != 0:
+                raise RuntimeError("csp_memcpy_d2h failed")
+            view = (ctypes.c_char * nbytes).from_address(dst)
+            stats["pinned_tensors"] += 1
+            stats["mode"] = "pinned"
+            return emit_buffer(view, None, src_t.dtype, src_t.shape)
+        # large CPU tensor: zero-copy out-of-band (keep tensor alive)
+        src_t = t.contiguous()
+        view = (ctypes.c_char * nbytes).from_address(src_t.data_ptr())
+        if stats["mode"] != "pinned":
+            stats["mode"] = "cpu-oob"
+        return emit_buffer(view, src_t, src_t.dtype, src_t.shape)
 
     def walk(obj):
         if isinstance(obj, torch.Tensor):
-            return to_host(obj) if obj.is_cuda else obj
+            return to_host(obj)
         if isinstance(obj, dict):
             return {k: walk(v) for k, v in obj.items()}
         if isinstance(obj, tuple):
@@ -212,11 +241,12 @@ def _write_frame(fd, payload):
 def _serve_one(request):
     t0 = time.monotonic()
     meta = {"phases_ms": {}, "gpu": _gpu_info, "staging": None,
-            "gpu_slot": GPU_SLOT,
+            "gpu_slot": GPU_SLOT, "buffers": [],
             "hip_visible_devices": os.environ.get("HIP_VISIBLE_DEVICES"),
             "pid": os.getpid(), "worker": True}
     result = None
     exception = None
+    buffers = []
     try:
         fn, args, kwargs = pickle.loads(request["function_blob"])
     except Exception as e:  # noqa: BLE001
@@ -245,18 +275,21 @@ def _serve_one(request):
         t_stage = time.monotonic()
         stats = {"tensors": 0, "pinned_tensors": 0, "bytes": 0, "mode": "none"}
         try:
-            result = _stage_result(result, stats)
+            result = _stage_result(result, stats, buffers, meta["buffers"])
             meta["staging"] = stats
         except Exception as e:  # noqa: BLE001
-            result, exception = None, e
+            result, exception, buffers = None, e, []
+            meta["buffers"] = []
         meta["phases_ms"]["staging"] = round((time.monotonic() - t_stage) * 1000, 3)
 
     try:
         result_blob = pickle.dumps((result, exception))
     except Exception as e:  # noqa: BLE001
         result_blob = pickle.dumps((None, e))
+        buffers = []
+        meta["buffers"] = []
     meta["phases_ms"]["total"] = round((time.monotonic() - t0) * 1000, 3)
-    return result_blob, meta
+    return result_blob, meta, buffers
 
 
 def main():
@@ -278,8 +311,13 @@ def main():
         if frame is None or frame == b"":
             break
         request = pickle.loads(frame)
-        result_blob, meta = _serve_one(request)
-        _write_frame(_proto_fd, pickle.dumps((result_blob, meta)))
+        result_blob, meta, buffers = _serve_one(request)
+        _write_frame(_proto_fd, pickle.dumps(("R1", result_blob, meta, len(buffers))))
+        for view, _keep in buffers:
+            _write_frame(_proto_fd, view)
+        del buffers
+        if _gpu_lib is not None:
+            _gpu_lib.csp_staging_release_all()
 
 
 if __name__ == "__main__":

@@ -90,6 +90,38 @@ async def get_worker(key: WorkerKey, launcher, startup_timeout: float = 180.0) -
         return handle
 
 
+def _reconstruct(result, buffer_meta, buffers):
+    """Replace out-of-band tensor-buffer markers with rebuilt torch
+    tensors (workers ship large tensors as raw frames; see
+    remote/worker_template.py protocol note)."""
+    import torch
+
+    def build(i: int):
+        info = buffer_meta[i]
+        dtype = getattr(torch, info["dtype"])
+        data = bytearray(buffers[i])  # writable for torch.frombuffer
+        t = torch.frombuffer(data, dtype=dtype)
+        return t.reshape(info["shape"])
+
+    def walk(obj):
+        if (
+            isinstance(obj, tuple)
+            and len(obj) == 2
+            and obj[0] == "__csp_tensor_buffer_v1__"
+        ):
+            return build(obj[1])
+        if isinstance(obj, dict):
+            return {k: walk(v) for k, v in obj.items()}
+        if isinstance(obj, tuple):
+            vals = [walk(v) for v in obj]
+            return type(obj)(*vals) if hasattr(obj, "_fields") else tuple(vals)
+        if isinstance(obj, list):
+            return [walk(v) for v in obj]
+        return obj
+
+    return walk(result)
+
+
 async def run_task(
     handle: WorkerHandle,
     op_id: str,
@@ -101,9 +133,15 @@ async def run_task(
     request = cloudpickle.dumps(
         {"op_id": op_id, "workdir": workdir, "function_blob": function_blob}
     )
-    reply = await handle.channel.request(request, timeout=timeout)
-    result_blob, meta = cloudpickle.loads(reply)
+    async with handle.channel.transaction() as ch:
+        await ch.send_frame(request)
+        main = await ch.recv_frame(timeout=timeout)
+        tag, result_blob, meta, nbuf = cloudpickle.loads(main)
+        assert tag == "R1", f"unexpected worker response tag {tag!r}"
+        buffers = [await ch.recv_frame(timeout=timeout) for _ in range(nbuf)]
     result, exception = cloudpickle.loads(result_blob)
+    if nbuf:
+        result = _reconstruct(result, meta.get("buffers", []), buffers)
     return result, exception, meta
 
 

@@ -504,7 +504,10 @@ class SSHExecutor(RemoteExecutor):
         from .transport.channel import ChannelClosed
 
         script_remote = await self._provision_worker_script(transport, gpu_lib)
-        key = (self._pool_key(), worker_tag)
+        # key includes the script identity so config changes (which ship a
+        # new content-addressed script) spawn fresh workers instead of
+        # talking a new protocol to a stale process
+        key = (self._pool_key(), worker_tag, script_remote)
         cmd = self._wrap_conda(
             f"{self.python_path} {shlex.quote(script_remote)}"
         )

@@ -53,10 +53,16 @@ class Channel:
             raise ChannelClosed(f"{self._label}: EOF mid-frame") from e
 
     async def request(self, payload: bytes, timeout: Optional[float] = None) -> bytes:
-        """Serialized request/response round trip."""
+        """Serialized request/response round trip (single response frame)."""
         async with self._lock:
             await self.send_frame(payload)
             return await self.recv_frame(timeout=timeout)
+
+    def transaction(self):
+        """Async context manager holding the channel's request lock for a
+        multi-frame exchange (e.g. a response followed by raw
+        tensor-buffer frames)."""
+        return _Transaction(self)
 
     def kill(self) -> None:
         """Synchronous hard-kill (for reaping workers whose event loop is
@@ -86,6 +92,18 @@ class Channel:
             except asyncio.TimeoutError:
                 self._proc.kill()
                 await self._proc.wait()
+
+
+class _Transaction:
+    def __init__(self, channel: "Channel"):
+        self._channel = channel
+
+    async def __aenter__(self) -> "Channel":
+        await self._channel._lock.acquire()
+        return self._channel
+
+    async def __aexit__(self, *exc) -> None:
+        self._channel._lock.release()
 
 
 async def open_subprocess_channel(argv: List[str], label: str) -> Channel:

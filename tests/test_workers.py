@@ -114,3 +114,48 @@ def test_worker_large_payload(local_executor):
 
     out = asyncio.run(ex.execute(big, [8 * 1024 * 1024], {}))
     assert len(out) == 8 * 1024 * 1024
+
+
+def test_worker_out_of_band_tensor_buffers(local_executor):
+    """Large CPU tensors return via raw out-of-band frames (the same path
+    CUDA tensors take through pinned staging on a GPU box)."""
+    torch = __import__("pytest").importorskip("torch")
+    ex = local_executor(
+        persistent_workers=True, pinned_staging_threshold_bytes=1024
+    )
+
+    def fn(n):
+        import torch
+
+        big = torch.arange(n, dtype=torch.float32).reshape(4, -1)
+        small = torch.ones(3, dtype=torch.bfloat16)
+        return {"big": big, "nested": [small, (big * 2, "tag")], "n": n}
+
+    out = asyncio.run(ex.execute(fn, [4096], {}, dispatch_id="oob", node_id=0))
+    assert out["n"] == 4096
+    assert torch.equal(
+        out["big"], torch.arange(4096, dtype=torch.float32).reshape(4, -1)
+    )
+    assert torch.equal(out["nested"][1][0], out["big"] * 2)
+    assert out["nested"][0].dtype == torch.bfloat16
+    meta = ex.last_task_record.remote_meta
+    assert meta["staging"]["mode"] == "cpu-oob"
+    assert len(meta["buffers"]) == 2  # the two large tensors
+    assert meta["staging"]["tensors"] == 3
+
+
+def test_worker_large_tensor_roundtrip_exact(local_executor):
+    torch = __import__("pytest").importorskip("torch")
+    ex = local_executor(
+        persistent_workers=True, pinned_staging_threshold_bytes=1 << 20
+    )
+
+    def fn():
+        import torch
+
+        g = torch.Generator().manual_seed(7)
+        return torch.randn(1024, 1024, generator=g)
+
+    out = asyncio.run(ex.execute(fn, [], {}))
+    g = torch.Generator().manual_seed(7)
+    assert torch.equal(out, torch.randn(1024, 1024, generator=g))
