@@ -69,6 +69,19 @@ _proto_fd = os.dup(1)
 os.dup2(2, 1)
 sys.stdout = sys.stderr
 
+# widen the protocol pipes (default 64 KiB caps large-tensor frames)
+try:
+    import fcntl
+
+    F_SETPIPE_SZ = 1031  # linux
+    for _fd in (0, _proto_fd):
+        try:
+            fcntl.fcntl(_fd, F_SETPIPE_SZ, 1 << 20)
+        except OSError:
+            pass
+except ImportError:
+    pass
+
 import cloudpickle as pickle  # noqa: E402  (after fd surgery on purpose)
 
 _gpu_lib = None

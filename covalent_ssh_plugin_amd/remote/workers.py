@@ -99,7 +99,10 @@ def _reconstruct(result, buffer_meta, buffers):
     def build(i: int):
         info = buffer_meta[i]
         dtype = getattr(torch, info["dtype"])
-        data = bytearray(buffers[i])  # writable for torch.frombuffer
+        raw = buffers[i]
+        # large frames arrive as a preallocated bytearray (zero-copy
+        # here); small ones as bytes (copy once into writable memory)
+        data = raw if isinstance(raw, bytearray) else bytearray(raw)
         t = torch.frombuffer(data, dtype=dtype)
         return t.reshape(info["shape"])
 

@@ -43,7 +43,21 @@ class Channel:
             (length,) = struct.unpack(">I", header)
             if length == 0:
                 return b""
-            return await self._proc.stdout.readexactly(length)
+            if length <= (4 << 20):
+                return await self._proc.stdout.readexactly(length)
+            # large frame (tensor buffer): assemble into ONE preallocated
+            # bytearray instead of readexactly's chunk-list + join + a
+            # later bytearray copy — tensors reconstruct zero-copy on it
+            buf = bytearray(length)
+            view = memoryview(buf)
+            pos = 0
+            while pos < length:
+                chunk = await self._proc.stdout.read(min(1 << 20, length - pos))
+                if not chunk:
+                    raise asyncio.IncompleteReadError(bytes(view[:pos]), length)
+                view[pos : pos + len(chunk)] = chunk
+                pos += len(chunk)
+            return buf
 
         try:
             if timeout is None:
