@@ -152,6 +152,15 @@ def drop(key: WorkerKey) -> None:
     _workers.pop(key, None)
 
 
+async def kill(key: WorkerKey) -> None:
+    """Hard-kill the worker for ``key`` (task cancellation): its
+    in-flight request fails with ChannelClosed; the next electron for
+    the key spawns a fresh worker."""
+    handle = _workers.pop(key, None)
+    if handle is not None:
+        handle.channel.kill()
+
+
 async def close_all() -> None:
     handles = list(_workers.values())
     _workers.clear()

@@ -185,6 +185,10 @@ class SSHExecutor(RemoteExecutor):
         self.fixed_gpu = int(fixed_gpu)
         self.local_home = local_home
 
+        #: in-flight worker-dispatched tasks: operation_id -> worker key
+        self._inflight: Dict[str, Tuple] = {}
+        #: operation_ids cancelled while in flight
+        self._cancelled: set = set()
         #: most recent completed task's per-phase timing record
         self.last_task_record: Optional[TaskRecord] = None
         #: bounded history of task records (bench reads this)
@@ -329,9 +333,29 @@ class SSHExecutor(RemoteExecutor):
         with open(local_result_file, "rb") as f:
             return stdlib_pickle.load(f)
 
-    async def cancel(self, *args: Any, **kwargs: Any) -> None:
-        """Unsupported, as in the reference (ssh.py:460-464)."""
-        raise NotImplementedError("Cancellation is not supported by the SSH executor")
+    async def cancel(self, task_metadata: Optional[dict] = None, *args: Any, **kwargs: Any) -> None:
+        """Cancel a running task.
+
+        The reference leaves this unimplemented (reference ssh.py:460-464).
+        Here, a task dispatched through a persistent worker CAN be
+        cancelled: the worker process serving it is killed (the in-flight
+        request fails with ChannelClosed and a fresh worker replaces it
+        for subsequent electrons).  Stub-dispatched tasks remain
+        uncancellable, matching the reference.
+        """
+        task_metadata = task_metadata or {}
+        operation_id = (
+            f"{task_metadata.get('dispatch_id', 'dispatch')}_"
+            f"{task_metadata.get('node_id', 0)}"
+        )
+        key = self._inflight.get(operation_id)
+        if key is None:
+            raise NotImplementedError(
+                "Cancellation is only supported for tasks running on "
+                f"persistent workers (no in-flight worker task {operation_id!r})"
+            )
+        self._cancelled.add(operation_id)
+        await worker_pool.kill(key)
 
     # ------------------------------------------------------------------
     # Staging (reference ssh.py:126-179)
@@ -515,19 +539,26 @@ class SSHExecutor(RemoteExecutor):
         async def launcher():
             return await transport.open_channel(cmd, env=env)
 
-        for attempt in (0, 1):
-            handle = await worker_pool.get_worker(key, launcher)
-            try:
-                return await worker_pool.run_task(
-                    handle, operation_id, workdir, function_blob
-                )
-            except ChannelClosed:
-                worker_pool.drop(key)
-                if attempt == 1:
-                    raise
-                app_log.warning(
-                    "worker %s died mid-task; respawning once", key
-                )
+        self._inflight[operation_id] = key
+        try:
+            for attempt in (0, 1):
+                handle = await worker_pool.get_worker(key, launcher)
+                try:
+                    return await worker_pool.run_task(
+                        handle, operation_id, workdir, function_blob
+                    )
+                except ChannelClosed:
+                    worker_pool.drop(key)
+                    if operation_id in self._cancelled:
+                        self._cancelled.discard(operation_id)
+                        raise SSHTaskError(f"task {operation_id} was cancelled")
+                    if attempt == 1:
+                        raise
+                    app_log.warning(
+                        "worker %s died mid-task; respawning once", key
+                    )
+        finally:
+            self._inflight.pop(operation_id, None)
 
     # ------------------------------------------------------------------
     # Fused single-round-trip dispatch

@@ -13,10 +13,11 @@ from pathlib import Path
 from setuptools import find_packages, setup
 
 here = Path(__file__).parent
+version = (here / "VERSION").read_text().strip()
 
 setup(
     name="covalent-ssh-plugin-amd",
-    version="0.1.0",
+    version=version,
     description=(
         "MI355X-native Covalent SSH executor: pooled multiplexed SSH dispatch "
         "onto an 8xMI355X node with per-task GPU slots, a CDNA4 warm-up/probe "

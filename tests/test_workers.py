@@ -160,3 +160,40 @@ def test_worker_large_tensor_roundtrip_exact(local_executor):
     out = asyncio.run(ex.execute(fn, [], {}))
     g = torch.Generator().manual_seed(7)
     assert torch.equal(out, torch.randn(1024, 1024, generator=g))
+
+
+def test_cancel_kills_worker_task(local_executor):
+    """cancel() on a worker-dispatched task kills the serving worker; the
+    task fails and the next electron gets a fresh worker (reference
+    parity: stub tasks stay uncancellable)."""
+    ex = local_executor(persistent_workers=True, cpu_workers=1)
+
+    def hang():
+        import time
+
+        time.sleep(60)
+        return "never"
+
+    async def main():
+        task = asyncio.ensure_future(
+            ex.execute(hang, [], {}, dispatch_id="cx", node_id=0)
+        )
+        # wait until the task is actually in flight on a worker
+        for _ in range(200):
+            if "cx_0" in ex._inflight:
+                await asyncio.sleep(0.2)  # let the request frame land
+                break
+            await asyncio.sleep(0.01)
+        await ex.cancel({"dispatch_id": "cx", "node_id": 0})
+        with pytest.raises(RuntimeError):
+            await task
+        # executor recovers: next electron runs on a fresh worker
+        return await ex.execute(lambda: 123, [], {}, dispatch_id="cx", node_id=1)
+
+    assert asyncio.run(main()) == 123
+
+
+def test_cancel_unknown_task_not_implemented(local_executor):
+    ex = local_executor()
+    with pytest.raises(NotImplementedError):
+        asyncio.run(ex.cancel({"dispatch_id": "nope", "node_id": 9}))
