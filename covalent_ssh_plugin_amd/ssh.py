@@ -542,7 +542,15 @@ class SSHExecutor(RemoteExecutor):
         self._inflight[operation_id] = key
         try:
             for attempt in (0, 1):
-                handle = await worker_pool.get_worker(key, launcher)
+                try:
+                    handle = await worker_pool.get_worker(key, launcher)
+                except worker_pool.WorkerStartupError:
+                    # transient spawn failure (loaded machine, dropped
+                    # channel): one clean retry before surfacing
+                    if attempt == 1:
+                        raise
+                    await asyncio.sleep(0.5)
+                    continue
                 try:
                     return await worker_pool.run_task(
                         handle, operation_id, workdir, function_blob
