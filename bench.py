@@ -91,8 +91,8 @@ def build_electron(config: str):
 def main() -> None:
     parser = argparse.ArgumentParser()
     parser.add_argument("--gpus", type=int, default=1)
-    parser.add_argument("--steps", type=int, default=32)
-    parser.add_argument("--warmup", type=int, default=8)
+    parser.add_argument("--steps", type=int, default=256)
+    parser.add_argument("--warmup", type=int, default=32)
     parser.add_argument("--config", default="noop")
     parser.add_argument("--fan", type=int, default=64,
                         help="concurrent electrons per step for --config fan")
