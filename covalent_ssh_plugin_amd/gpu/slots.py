@@ -20,8 +20,8 @@ across the node even at low concurrency.
 from __future__ import annotations
 
 import asyncio
-import itertools
-from dataclasses import dataclass, field
+
+from dataclasses import dataclass
 from typing import Dict, List, Optional, Tuple
 
 

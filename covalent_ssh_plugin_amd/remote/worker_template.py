@@ -31,7 +31,7 @@ re-pointed at stderr before any user code runs, so user prints cannot
 corrupt frames.
 """
 
-import io
+
 import json
 import os
 import struct

@@ -816,6 +816,14 @@ class SSHExecutor(RemoteExecutor):
         record.total_s = timer.total()
         self.last_task_record = record
         self.task_records.append(record)
+        log_path = os.environ.get("CSP_AMD_TASK_LOG")
+        if log_path:
+            # structured per-task trace (SURVEY.md §5 tracing row)
+            try:
+                with open(log_path, "a") as f:
+                    f.write(record.to_json() + "\n")
+            except OSError:
+                app_log.debug("task log write failed", exc_info=True)
         if len(self.task_records) > 10000:
             del self.task_records[: len(self.task_records) // 2]
 

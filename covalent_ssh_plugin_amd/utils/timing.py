@@ -58,6 +58,17 @@ class TaskRecord:
     def load_meta_file(self, path: str) -> None:
         self.load_meta(Path(path).read_bytes())
 
+    def to_json(self) -> str:
+        return json.dumps(
+            {
+                "operation_id": self.operation_id,
+                "gpu_id": self.gpu_id,
+                "total_s": round(self.total_s, 6),
+                "phases": {k: round(v, 6) for k, v in self.phases.items()},
+                "remote_meta": self.remote_meta,
+            }
+        )
+
 
 def summarize(records) -> dict:
     """Aggregate a list of TaskRecords into dispatch statistics

@@ -188,3 +188,19 @@ def test_stats_summary(local_executor):
     assert s["count"] == 4
     assert s["p50_ms"] > 0
     assert "dispatch" in s["phase_mean_ms"]
+
+
+def test_task_log_jsonl(local_executor, tmp_path, monkeypatch):
+    log = tmp_path / "tasks.jsonl"
+    monkeypatch.setenv("CSP_AMD_TASK_LOG", str(log))
+    ex = local_executor()
+
+    async def main():
+        for i in range(3):
+            await ex.execute(_add, [i, 1], {}, dispatch_id="log", node_id=i)
+
+    asyncio.run(main())
+    lines = [json.loads(l) for l in log.read_text().splitlines()]
+    assert len(lines) == 3
+    assert lines[0]["operation_id"] == "log_0"
+    assert "dispatch" in lines[0]["phases"]

@@ -151,3 +151,19 @@ def test_split_stream_sentinels():
     # result but no meta
     t, r, m = SSHExecutor._split_stream(task_out + s_result + result, s_result, s_meta)
     assert r == result and m is None
+
+
+def test_real_ssh_client_connect_refused(tmp_path):
+    """Runs the REAL ssh binary against a closed local port: the
+    transport must surface TransportConnectError (feeds the executor's
+    retry loop), not hang or crash."""
+    t = OpenSSHTransport(
+        hostname="127.0.0.1",
+        username="nobody",
+        port=1,  # reserved port, nothing listens
+        control_dir=str(tmp_path),
+        connect_timeout=5,
+    )
+    with pytest.raises(TransportConnectError):
+        asyncio.run(t.connect())
+    assert not t.is_connected
