@@ -165,17 +165,18 @@ static int run_mfma_spin(int iters_per_wave, int blocks, float* ms_out) {
 }
 
 static int run_hbm_sweep(float4* buf_a, float4* buf_b, size_t n4, int reps,
-                         float* ms_out, int variant = -1) {
+                         float* ms_out, int variant = -1, int blocks = 4096,
+                         bool pingpong = true) {
     sweep_fn kern = sweep_variant(variant < 0 ? g_sweep_variant : variant);
     hipEvent_t t0, t1;
     HIP_TRY(hipEventCreate(&t0));
     HIP_TRY(hipEventCreate(&t1));
     HIP_TRY(hipEventRecord(t0));
     for (int r = 0; r < reps; ++r) {
-        // ping-pong so reads are never served from a just-written line
-        const float4* s = (r & 1) ? buf_b : buf_a;
-        float4* d = (r & 1) ? buf_a : buf_b;
-        hipLaunchKernelGGL(kern, dim3(4096), dim3(256), 0, 0, s, d, n4);
+        // ping-pong alternates direction; fixed always copies a -> b
+        const float4* s = (pingpong && (r & 1)) ? buf_b : buf_a;
+        float4* d = (pingpong && (r & 1)) ? buf_a : buf_b;
+        hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), 0, 0, s, d, n4);
     }
     HIP_TRY(hipEventRecord(t1));
     HIP_TRY(hipEventSynchronize(t1));
@@ -392,6 +393,26 @@ extern "C" int csp_hbm_bench(int device, int variant, size_t bytes, int reps,
     float warm_ms = 0.f, ms = 0.f;
     int rc = run_hbm_sweep(a, b, n4, 2, &warm_ms, variant);
     if (rc == 0) rc = run_hbm_sweep(a, b, n4, reps, &ms, variant);
+    (void)hipFree(a);
+    (void)hipFree(b);
+    if (rc != 0) return rc;
+    *gbps = (2.0 * (double)bytes * reps / 1.0e9) / ((double)ms / 1000.0);
+    return 0;
+}
+
+// Extended exploration: grid size + copy direction knobs.
+extern "C" int csp_hbm_bench2(int device, int variant, size_t bytes, int reps,
+                              int blocks, int pingpong, double* gbps) {
+    HIP_TRY(hipSetDevice(device));
+    size_t n4 = bytes / sizeof(float4);
+    float4 *a = nullptr, *b = nullptr;
+    HIP_TRY(hipMalloc(&a, bytes));
+    HIP_TRY(hipMalloc(&b, bytes));
+    HIP_TRY(hipMemsetAsync(a, 1, bytes));
+    float warm_ms = 0.f, ms = 0.f;
+    int rc = run_hbm_sweep(a, b, n4, 2, &warm_ms, variant, blocks, pingpong != 0);
+    if (rc == 0)
+        rc = run_hbm_sweep(a, b, n4, reps, &ms, variant, blocks, pingpong != 0);
     (void)hipFree(a);
     (void)hipFree(b);
     if (rc != 0) return rc;
