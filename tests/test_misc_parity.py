@@ -59,3 +59,17 @@ def test_sdist_builds(tmp_path):
     assert any(n.endswith("ops/hip/csp_gpu.hip") for n in names)
     assert any(n.endswith("remote/stub_template.py") for n in names)
     assert any(n.endswith("remote/worker_template.py") for n in names)
+
+
+def test_executor_instance_is_picklable():
+    """Covalent serializes executor instances into its dispatch records;
+    the executor must survive a pickle round trip (transports/workers are
+    module-level pooled state, not instance state)."""
+    import pickle
+
+    ex = SSHExecutor(username="u", hostname="h", gpu_slots=4, persistent_workers=True)
+    ex2 = pickle.loads(pickle.dumps(ex))
+    assert ex2.username == "u"
+    assert ex2.hostname == "h"
+    assert ex2.gpu_slots == 4
+    assert ex2.persistent_workers is True
