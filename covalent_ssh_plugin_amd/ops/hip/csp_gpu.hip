@@ -421,3 +421,15 @@ extern "C" int csp_hbm_bench2(int device, int variant, size_t bytes, int reps,
     *gbps = (2.0 * (double)bytes * reps / 1.0e9) / ((double)ms / 1000.0);
     return 0;
 }
+
+// MFMA spin exploration: blocks knob (TF/s out).
+extern "C" int csp_mfma_bench(int device, int iters, int blocks, double* tflops) {
+    HIP_TRY(hipSetDevice(device));
+    float ms = 0.f;
+    int rc = run_mfma_spin(iters / 4, blocks, &ms);  // warm clocks
+    if (rc == 0) rc = run_mfma_spin(iters, blocks, &ms);
+    if (rc != 0) return rc;
+    double waves = (double)blocks * 4.0;
+    *tflops = waves * (double)iters * kFlopPerMfma / ((double)ms / 1000.0) / 1.0e12;
+    return 0;
+}
