@@ -242,7 +242,7 @@ extern "C" int csp_probe_json(int device, char* buf, size_t buflen) {
     double hbm_gbps = (2.0 * (double)bytes * 6.0 / 1.0e9) / ((double)ms / 1000.0);
 
     // --- measured bf16 MFMA throughput ---------------------------------
-    const int blocks = 1024;  // 4x CU count: fills all 8 XCDs
+    const int blocks = 2048;  // 8 WGs/CU; best measured (profiles/)
     const int iters = 8192;   // per wave
     rc = run_mfma_spin(iters / 4, blocks, &warm_ms);  // warm clocks
     if (rc != 0) return rc;
