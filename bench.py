@@ -94,6 +94,8 @@ def main() -> None:
     parser.add_argument("--steps", type=int, default=256)
     parser.add_argument("--warmup", type=int, default=32)
     parser.add_argument("--config", default="noop")
+    parser.add_argument("--dump-latencies", default="",
+                        help="write per-step latencies (seconds, one per line)")
     parser.add_argument("--fan", type=int, default=64,
                         help="concurrent electrons per step for --config fan")
     args = parser.parse_args()
@@ -110,7 +112,11 @@ def main() -> None:
         import torch.distributed as dist
 
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        backend = "nccl" if has_cuda else "gloo"
+        # nccl needs one distinct device per rank; fall back to gloo when
+        # ranks outnumber visible GPUs (degenerate test topologies)
+        backend = (
+            "nccl" if has_cuda and world_size <= torch.cuda.device_count() else "gloo"
+        )
         dist.init_process_group(backend=backend)
         if has_cuda:
             torch.cuda.set_device(local_rank)
@@ -191,6 +197,10 @@ def main() -> None:
         elapsed = float(t.item())
 
     electrons_per_step = args.fan if args.config == "fan" else 1
+    if args.dump_latencies and rank == 0:
+        with open(args.dump_latencies, "w") as f:
+            f.write("\n".join(f"{x:.9f}" for x in out["latencies"]))
+
     total_electrons = args.steps * world_size * electrons_per_step
     eps = total_electrons / elapsed
     p50_ms = statistics.median(out["latencies"]) * 1000.0

@@ -9,11 +9,17 @@ meta/timings file and staging policy.
 
 from __future__ import annotations
 
+import functools
 from pathlib import Path
 from typing import Optional
 
 _TEMPLATE_PATH = Path(__file__).parent / "stub_template.py"
 _WORKER_TEMPLATE_PATH = Path(__file__).parent / "worker_template.py"
+
+
+@functools.lru_cache(maxsize=2)
+def _template_text(path: str) -> str:
+    return Path(path).read_text()
 
 # 64 MiB: below this a plain torch .cpu() copy is cheap; above it the
 # hipHostMalloc-pinned D2H path wins (PCIe Gen5 x16 ~63 GB/s vs pageable).
@@ -31,7 +37,7 @@ def render_stub(
     staging_threshold: Optional[int] = None,
 ) -> str:
     """Return the per-task stub script text."""
-    text = _TEMPLATE_PATH.read_text()
+    text = _template_text(str(_TEMPLATE_PATH))
     thr = DEFAULT_STAGING_THRESHOLD if staging_threshold is None else int(staging_threshold)
     replacements = {
         "__CSP_RESULT_FILE__": remote_result_file,
@@ -47,6 +53,7 @@ def render_stub(
     return text
 
 
+@functools.lru_cache(maxsize=64)
 def render_worker(
     *,
     gpu_lib_path: str = "",
@@ -55,7 +62,7 @@ def render_worker(
 ) -> str:
     """Return the persistent worker script text (one per endpoint; the
     GPU slot arrives via the CSP_GPU_SLOT env var at launch)."""
-    text = _WORKER_TEMPLATE_PATH.read_text()
+    text = _template_text(str(_WORKER_TEMPLATE_PATH))
     thr = DEFAULT_STAGING_THRESHOLD if staging_threshold is None else int(staging_threshold)
     replacements = {
         "__CSP_GPU_LIB__": gpu_lib_path,

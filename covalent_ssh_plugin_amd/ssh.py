@@ -102,6 +102,16 @@ def _conf(key: str, explicit: Any, default: Any = None) -> Any:
     return value
 
 
+import functools
+
+
+@functools.lru_cache(maxsize=64)
+def _script_digest(text: str) -> str:
+    import hashlib
+
+    return hashlib.sha256(text.encode()).hexdigest()[:12]
+
+
 class SSHTaskError(RuntimeError):
     """Dispatcher-side failure of the SSH pipeline (not the user task)."""
 
@@ -486,7 +496,7 @@ class SSHExecutor(RemoteExecutor):
             warmup=self.warmup_gpu,
             staging_threshold=self.pinned_staging_threshold_bytes,
         )
-        digest = hashlib.sha256(text.encode()).hexdigest()[:12]
+        digest = _script_digest(text)
         key = self._pool_key()
         check_name = f"worker_script:{digest}"
         cached = transport_pool.cached_check(key, check_name)

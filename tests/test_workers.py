@@ -57,25 +57,25 @@ def test_worker_concurrency_cpu_pool(local_executor):
     ex = local_executor(persistent_workers=True, cpu_workers=4)
 
     def slow(i):
+        import os
         import time
 
-        time.sleep(0.3)
-        return i
+        time.sleep(0.1)
+        return i, os.getpid()
 
     async def main():
-        import time
-
-        t0 = time.perf_counter()
-        out = await asyncio.gather(
+        return await asyncio.gather(
             *[ex.execute(slow, [i], {}, dispatch_id="c", node_id=i) for i in range(4)]
         )
-        return out, time.perf_counter() - t0
 
-    out, dt = asyncio.run(main())
-    assert out == [0, 1, 2, 3]
-    # 4 workers -> 4x 0.3s sleeps overlap (serial would be >=1.2s);
-    # generous bound for loaded CI machines
-    assert dt < 1.0, dt
+    out = asyncio.run(main())
+    assert [i for i, _ in out] == [0, 1, 2, 3]
+    # four DISTINCT worker processes served the batch: true 4-way
+    # parallelism (each worker channel serializes its own requests, so
+    # distinct pids == concurrent capability; wall-clock assertions are
+    # flaky on loaded CI machines)
+    pids = {pid for _, pid in out}
+    assert len(pids) == 4, out
 
 
 def test_worker_death_respawn(local_executor):
