@@ -262,8 +262,13 @@ def main():
             result, exception = None, e
         _mark("staging", t_stage)
 
+    blob = pickle.dumps((result, exception))
     with open(RESULT_FILE, "wb") as f_out:
-        pickle.dump((result, exception), f_out)
+        f_out.write(blob)
+    # integrity check for the sentinel-framed fused return path
+    import hashlib
+
+    _meta["result_sha256"] = hashlib.sha256(blob).hexdigest()
     _write_meta()
 
 

@@ -746,9 +746,16 @@ class SSHExecutor(RemoteExecutor):
                     )
 
             if not self.persistent_workers and self.batch_roundtrips:
-                with timer.phase("dispatch"):
-                    proc, result_bytes, meta_bytes = await self._dispatch_fused(
-                        transport, paths, env
+                try:
+                    with timer.phase("dispatch"):
+                        proc, result_bytes, meta_bytes = await self._dispatch_fused(
+                            transport, paths, env
+                        )
+                except (TransportConnectError, OSError) as e:
+                    self._cleanup_local(paths)
+                    return await self._on_ssh_fail(
+                        function, args, kwargs,
+                        f"transport failed during task {operation_id}: {e}",
                     )
                 if proc.returncode != 0 or result_bytes is None:
                     message = (
@@ -758,9 +765,21 @@ class SSHExecutor(RemoteExecutor):
                     self._cleanup_local(paths)
                     return await self._on_ssh_fail(function, args, kwargs, message)
                 with timer.phase("fetch"):
-                    result, exception = stdlib_pickle.loads(result_bytes)
                     if meta_bytes:
                         record.load_meta(meta_bytes)
+                    expected = (record.remote_meta or {}).get("result_sha256")
+                    if expected:
+                        import hashlib
+
+                        actual = hashlib.sha256(result_bytes).hexdigest()
+                        if actual != expected:
+                            self._cleanup_local(paths)
+                            return await self._on_ssh_fail(
+                                function, args, kwargs,
+                                f"result stream for {operation_id} failed its "
+                                f"integrity check ({actual[:12]} != {expected[:12]})",
+                            )
+                    result, exception = stdlib_pickle.loads(result_bytes)
                     if self.do_cleanup is False:
                         Path(paths["result_local"]).write_bytes(result_bytes)
             elif not self.persistent_workers:

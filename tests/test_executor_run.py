@@ -36,6 +36,8 @@ def test_fused_roundtrip(local_executor):
     assert rec.operation_id == "d_0"
     assert rec.phases["dispatch"] > 0
     assert rec.remote_meta is not None and "total" in rec.remote_meta["phases_ms"]
+    # every fused return is integrity-checked against the stub's hash
+    assert len(rec.remote_meta["result_sha256"]) == 64
 
 
 def test_template_path_roundtrip(local_executor):
