@@ -94,3 +94,42 @@ def test_tar_stream_mixed_rejected(tmp_path):
     f.write_bytes(b"abc")
     with pytest.raises(ValueError):
         make_tar_stream([(str(f), "/abs/a.bin"), (str(f), "rel/a.bin")])
+
+
+def test_put_files_empty_batch(tmp_path):
+    async def main():
+        t = LocalTransport(home=str(tmp_path))
+        await t.connect()
+        await t.put_files([])  # no-op, must not raise
+
+    asyncio.run(main())
+
+
+def test_summarize_empty_records():
+    from covalent_ssh_plugin_amd.utils.timing import summarize
+
+    assert summarize([]) == {"count": 0}
+
+
+def test_pool_reconnects_dropped_transport(tmp_path):
+    from covalent_ssh_plugin_amd.transport import pool as transport_pool
+
+    async def main():
+        home = tmp_path / "h"
+        home.mkdir()
+        made = []
+
+        def factory():
+            t = LocalTransport(home=str(home))
+            made.append(t)
+            return t
+
+        t1 = await transport_pool.get_transport(("k",), factory)
+        assert t1.is_connected
+        await t1.close()  # connection dropped
+        t2 = await transport_pool.get_transport(("k",), factory)
+        assert t2 is t1  # same pooled instance...
+        assert t2.is_connected  # ...reconnected
+        assert len(made) == 1
+
+    asyncio.run(main())
