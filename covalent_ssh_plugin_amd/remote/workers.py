@@ -61,6 +61,19 @@ def cpu_worker_index(pool_size: int) -> int:
     return idx
 
 
+def pick_cpu_tag(pool_key, pool_size: int) -> str:
+    """Pick a CPU worker tag: prefer an already-running IDLE worker
+    (avoids head-of-line blocking behind a long task), else round-robin
+    (which also spreads initial spawns across the set)."""
+    candidates = [f"cpu{i}" for i in range(max(1, pool_size))]
+    for tag in candidates:
+        for key, handle in _workers.items():
+            if len(key) >= 2 and key[0] == pool_key and key[1] == tag:
+                if handle.alive and not handle.channel._lock.locked():
+                    return tag
+    return f"cpu{cpu_worker_index(pool_size)}"
+
+
 async def get_worker(key: WorkerKey, launcher, startup_timeout: float = 180.0) -> WorkerHandle:
     """Return the live worker for ``key``, starting it with ``launcher``
     (an async callable returning a Channel) if needed."""

@@ -731,7 +731,9 @@ class SSHExecutor(RemoteExecutor):
                             (
                                 record.gpu_id
                                 if record.gpu_id is not None
-                                else f"cpu{worker_pool.cpu_worker_index(self.cpu_workers)}"
+                                else worker_pool.pick_cpu_tag(
+                                    self._pool_key(), self.cpu_workers
+                                )
                             ),
                         )
                     record.remote_meta = meta
@@ -866,6 +868,45 @@ class SSHExecutor(RemoteExecutor):
                 os.remove(paths[key])
             except OSError:
                 pass
+
+    async def prewarm(self, slots: Optional[int] = None) -> int:
+        """Spin up transports, env checks and persistent workers (with
+        their GPU prologue) ahead of the first electron, so first-task
+        latency matches steady state.  Returns the number of workers
+        started.  No-op unless ``persistent_workers`` is enabled."""
+        await self._validate_credentials()
+        transport = await self._client_connect()
+        if transport is None:
+            raise RuntimeError(f"could not connect to {self.hostname}")
+        gpu_lib = await self._ensure_environment(transport)
+        if not self.persistent_workers:
+            return 0
+        script_remote = await self._provision_worker_script(transport, gpu_lib)
+        cmd = self._wrap_conda(f"{self.python_path} {shlex.quote(script_remote)}")
+        started = 0
+        if self.hip_visible_devices_policy == "roundrobin" and gpu_lib:
+            tags_envs = [
+                (gpu, {"CSP_GPU_SLOT": str(gpu)})
+                for gpu in range(slots if slots is not None else self.gpu_slots)
+            ]
+        elif self.hip_visible_devices_policy == "fixed" and gpu_lib:
+            tags_envs = [(self.fixed_gpu, {"CSP_GPU_SLOT": str(self.fixed_gpu)})]
+        else:
+            tags_envs = [
+                (f"cpu{i}", None)
+                for i in range(slots if slots is not None else self.cpu_workers)
+            ]
+
+        async def start(tag, env):
+            key = (self._pool_key(), tag, script_remote)
+
+            async def launcher():
+                return await transport.open_channel(cmd, env=env)
+
+            await worker_pool.get_worker(key, launcher)
+
+        await asyncio.gather(*[start(t, e) for t, e in tags_envs])
+        return len(tags_envs)
 
     def stats(self) -> dict:
         """Latency percentiles + per-phase means over this executor's

@@ -197,3 +197,53 @@ def test_cancel_unknown_task_not_implemented(local_executor):
     ex = local_executor()
     with pytest.raises(NotImplementedError):
         asyncio.run(ex.cancel({"dispatch_id": "nope", "node_id": 9}))
+
+
+def test_prewarm_starts_workers(local_executor):
+    ex = local_executor(persistent_workers=True, cpu_workers=3)
+
+    async def main():
+        n = await ex.prewarm()
+        assert n == 3
+        # electrons reuse prewarmed workers (no spawn in the timed path)
+        import time
+
+        t0 = time.perf_counter()
+        out = await ex.execute(lambda: 1, [], {})
+        dt = time.perf_counter() - t0
+        return out, dt
+
+    out, dt = asyncio.run(main())
+    assert out == 1
+
+
+def test_pick_cpu_tag_prefers_idle(local_executor):
+    """A long task must not block a following short task behind the same
+    worker: the short task picks a different (idle or fresh) worker."""
+    ex = local_executor(persistent_workers=True, cpu_workers=2)
+
+    def slow():
+        import time
+
+        time.sleep(1.0)
+        return "slow"
+
+    def quick():
+        return "quick"
+
+    async def main():
+        slow_task = asyncio.ensure_future(
+            ex.execute(slow, [], {}, dispatch_id="p", node_id=0)
+        )
+        await asyncio.sleep(0.2)  # slow task is in flight on one worker
+        import time
+
+        t0 = time.perf_counter()
+        q = await ex.execute(quick, [], {}, dispatch_id="p", node_id=1)
+        dt = time.perf_counter() - t0
+        s = await slow_task
+        return q, s, dt
+
+    q, s, dt = asyncio.run(main())
+    assert (q, s) == ("quick", "slow")
+    assert dt < 0.7, f"short task was blocked behind the long one ({dt:.2f}s)"
