@@ -277,11 +277,24 @@ def _serve_one(request):
 
     if exception is None:
         t_fn = time.monotonic()
+        # capture the task's python-level stdout/stderr into the meta
+        # (fd-level output still flows to the worker's stderr stream)
+        import contextlib
+        import io as _io
+
+        out_buf, err_buf = _io.StringIO(), _io.StringIO()
         try:
-            result = fn(*args, **kwargs)
+            with contextlib.redirect_stdout(out_buf), contextlib.redirect_stderr(err_buf):
+                result = fn(*args, **kwargs)
         except Exception as e:  # noqa: BLE001
             exception = e
         meta["phases_ms"]["user_fn"] = round((time.monotonic() - t_fn) * 1000, 3)
+        meta["stdout"] = out_buf.getvalue()[-65536:]
+        meta["stderr"] = err_buf.getvalue()[-65536:]
+        if meta["stdout"]:
+            sys.stderr.write(meta["stdout"])
+        if meta["stderr"]:
+            sys.stderr.write(meta["stderr"])
     os.chdir(home)
 
     if exception is None and result is not None:

@@ -247,3 +247,20 @@ def test_pick_cpu_tag_prefers_idle(local_executor):
     q, s, dt = asyncio.run(main())
     assert (q, s) == ("quick", "slow")
     assert dt < 0.7, f"short task was blocked behind the long one ({dt:.2f}s)"
+
+
+def test_worker_captures_task_output(local_executor):
+    """Per-task stdout/stderr land in the task record's remote meta."""
+    ex = local_executor(persistent_workers=True)
+
+    def chatty():
+        import sys
+
+        print("captured stdout line")
+        sys.stderr.write("captured stderr line\n")
+        return 1
+
+    assert asyncio.run(ex.execute(chatty, [], {})) == 1
+    meta = ex.last_task_record.remote_meta
+    assert "captured stdout line" in meta["stdout"]
+    assert "captured stderr line" in meta["stderr"]
