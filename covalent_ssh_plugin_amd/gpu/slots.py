@@ -3,10 +3,10 @@
 New component with no reference counterpart: the
 reference executes tasks on the remote with whatever GPU visibility the
 login shell has.  Here every task acquires a *slot* before it is
-submitted, and the remote command is launched with
-``HIP_VISIBLE_DEVICES=<gpu>`` (plus ``ROCR_VISIBLE_DEVICES`` for
-runtimes that consult it) so concurrent electrons land on distinct
-MI355X GPUs.  With 288 GB HBM3E per GPU there is no reason to co-locate
+submitted, and the remote command is launched with ``CSP_GPU_SLOT=<gpu>``,
+which the remote stub/worker resolves to ``HIP_VISIBLE_DEVICES`` within
+the host's ambient visibility list, so concurrent electrons land on
+distinct MI355X GPUs without clobbering pod GPU isolation.  With 288 GB HBM3E per GPU there is no reason to co-locate
 two electrons on one device; a slot is exactly one GPU by default
 (``slots_per_gpu`` can oversubscribe for small tasks).
 

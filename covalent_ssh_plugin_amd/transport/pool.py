@@ -41,6 +41,21 @@ def _lock() -> asyncio.Lock:
     return _pool_lock
 
 
+_check_locks: Dict[Tuple[PoolKey, object], Tuple[asyncio.Lock, object]] = {}
+
+
+def check_lock(key: PoolKey) -> asyncio.Lock:
+    """Loop-aware per-endpoint lock serializing one-time environment
+    checks (a cold-start fan of N electrons must run them once, not N
+    times)."""
+    loop = asyncio.get_running_loop()
+    entry = _check_locks.get(key)
+    if entry is None or entry[1] is not loop:
+        entry = (asyncio.Lock(), loop)
+        _check_locks[key] = entry
+    return entry[0]
+
+
 async def get_transport(key: PoolKey, factory: Callable[[], Transport]) -> Transport:
     """Return the pooled, connected transport for ``key``, creating it
     with ``factory`` (and connecting it) on first use.
@@ -70,6 +85,7 @@ async def close_all() -> None:
         transports = list(_pool.values())
         _pool.clear()
         _env_checks.clear()
+        _check_locks.clear()
     for t in transports:
         try:
             await t.close()
@@ -81,3 +97,4 @@ def reset() -> None:
     """Synchronous test hook: forget pooled state without closing."""
     _pool.clear()
     _env_checks.clear()
+    _check_locks.clear()

@@ -206,3 +206,26 @@ def test_task_log_jsonl(local_executor, tmp_path, monkeypatch):
     assert len(lines) == 3
     assert lines[0]["operation_id"] == "log_0"
     assert "dispatch" in lines[0]["phases"]
+
+
+def test_cold_start_env_checks_run_once(local_executor):
+    """A cold-start fan of electrons must run the per-endpoint env checks
+    once, not once per electron."""
+    ex = local_executor()
+    calls = []
+    orig = ex._ensure_environment_locked
+
+    async def counting(transport, key):
+        calls.append(1)
+        return await orig(transport, key)
+
+    ex._ensure_environment_locked = counting
+
+    async def main():
+        return await asyncio.gather(
+            *[ex.execute(_add, [i, 1], {}, dispatch_id="cold", node_id=i) for i in range(8)]
+        )
+
+    out = asyncio.run(main())
+    assert out == [i + 1 for i in range(8)]
+    assert len(calls) == 1, calls
