@@ -429,16 +429,15 @@ class SSHExecutor(RemoteExecutor):
             return gpu_lib
 
         async with transport_pool.check_lock(key):
+            cached = transport_pool.cached_check(key, "env")
+            if cached is not None:
+                ok, detail, gpu_lib = cached
+                if not ok:
+                    raise SSHTaskError(detail)
+                return gpu_lib
             return await self._ensure_environment_locked(transport, key)
 
     async def _ensure_environment_locked(self, transport: Transport, key) -> str:
-        cached = transport_pool.cached_check(key, "env")
-        if cached is not None:
-            ok, detail, gpu_lib = cached
-            if not ok:
-                raise SSHTaskError(detail)
-            return gpu_lib
-
         # conda env existence (reference ssh.py:508-519)
         if self.conda_env:
             proc = await transport.run(
