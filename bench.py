@@ -102,6 +102,9 @@ def main() -> None:
 
     import torch
 
+    # dmabuf IPC is required for RCCL on this host driver generation
+    os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
