@@ -16,7 +16,10 @@ python + HIP-runtime start per electron (the reference's architecture,
     result file) plus a meta JSON dict.
 
 Protocol (all frames: 4-byte big-endian length + payload):
-  request  = pickle dict {"op_id", "workdir", "function_blob"}
+  request  = pickle dict {"op_id", "workdir", "function_blob",
+             "arg_buffers": [{dtype, shape}, ...]} followed by one RAW
+             frame per entry (large CPU-tensor arguments, mirrored from
+             the dispatcher without pickle copies)
   response = pickle tuple ("R1", result_blob: bytes, meta: dict, nbuf: int)
              followed by nbuf RAW tensor-buffer frames.  Large tensors in
              the result are replaced by ("__csp_tensor_buffer_v1__", i)
@@ -251,6 +254,30 @@ def _write_frame(fd, payload):
         view = view[written:]
 
 
+def _rebuild_arg_tensors(obj, buffers, meta_list):
+    """Mirror of the dispatcher-side marker walk for request buffers."""
+    import torch
+
+    def build(i):
+        info = meta_list[i]
+        dtype = getattr(torch, info["dtype"])
+        return torch.frombuffer(buffers[i], dtype=dtype).reshape(info["shape"])
+
+    def walk(o):
+        if isinstance(o, tuple) and len(o) == 2 and o[0] == "__csp_tensor_buffer_v1__":
+            return build(o[1])
+        if isinstance(o, dict):
+            return {k: walk(v) for k, v in o.items()}
+        if isinstance(o, tuple):
+            vals = [walk(v) for v in o]
+            return type(o)(*vals) if hasattr(o, "_fields") else tuple(vals)
+        if isinstance(o, list):
+            return [walk(v) for v in o]
+        return o
+
+    return walk(obj)
+
+
 def _serve_one(request):
     t0 = time.monotonic()
     meta = {"phases_ms": {}, "gpu": _gpu_info, "staging": None,
@@ -262,6 +289,10 @@ def _serve_one(request):
     buffers = []
     try:
         fn, args, kwargs = pickle.loads(request["function_blob"])
+        arg_bufs = request.get("_arg_buffer_frames") or []
+        if arg_bufs:
+            args = _rebuild_arg_tensors(args, arg_bufs, request["arg_buffers"])
+            kwargs = _rebuild_arg_tensors(kwargs, arg_bufs, request["arg_buffers"])
     except Exception as e:  # noqa: BLE001
         exception = e
         fn = None
@@ -337,6 +368,15 @@ def main():
         if frame is None or frame == b"":
             break
         request = pickle.loads(frame)
+        n_arg_bufs = len(request.get("arg_buffers") or [])
+        if n_arg_bufs:
+            frames = []
+            for _ in range(n_arg_bufs):
+                f = _read_frame(0)
+                if f is None:
+                    return
+                frames.append(bytearray(f))
+            request["_arg_buffer_frames"] = frames
         result_blob, meta, buffers = _serve_one(request)
         _write_frame(_proto_fd, pickle.dumps(("R1", result_blob, meta, len(buffers))))
         for view, _keep in buffers:

@@ -103,6 +103,55 @@ async def get_worker(key: WorkerKey, launcher, startup_timeout: float = 180.0) -
         return handle
 
 
+def extract_arg_buffers(args, kwargs, threshold: int):
+    """Pull large contiguous CPU torch tensors out of (args, kwargs) and
+    replace them with out-of-band buffer markers (the request-direction
+    mirror of the worker's result staging).  Returns
+    (new_args, new_kwargs, buffer_meta, buffers); no-op (and no torch
+    import) unless torch is already loaded in the dispatcher."""
+    import sys
+
+    if "torch" not in sys.modules:
+        return args, kwargs, [], []
+    import ctypes
+
+    import torch
+
+    buffer_meta = []
+    buffers = []
+
+    def extract(t):
+        src_t = t.contiguous()
+        nbytes = src_t.numel() * src_t.element_size()
+        view = (ctypes.c_char * nbytes).from_address(src_t.data_ptr())
+        buffers.append((view, src_t))  # keep tensor alive until sent
+        buffer_meta.append(
+            {"dtype": str(src_t.dtype).replace("torch.", ""), "shape": list(src_t.shape)}
+        )
+        return ("__csp_tensor_buffer_v1__", len(buffers) - 1)
+
+    def walk(obj):
+        if isinstance(obj, torch.Tensor):
+            if (
+                not obj.is_cuda
+                and obj.numel() * obj.element_size() >= threshold
+            ):
+                return extract(obj)
+            return obj
+        if isinstance(obj, dict):
+            return {k: walk(v) for k, v in obj.items()}
+        if isinstance(obj, tuple):
+            vals = [walk(v) for v in obj]
+            return type(obj)(*vals) if hasattr(obj, "_fields") else tuple(vals)
+        if isinstance(obj, list):
+            return [walk(v) for v in obj]
+        return obj
+
+    new_args = walk(list(args))
+    new_kwargs = walk(dict(kwargs))
+    return new_args, new_kwargs, buffer_meta, buffers
+
+
 def _reconstruct(result, buffer_meta, buffers):
     """Replace out-of-band tensor-buffer markers with rebuilt torch
     tensors (workers ship large tensors as raw frames; see
@@ -144,13 +193,22 @@ async def run_task(
     workdir: str,
     function_blob: bytes,
     timeout: Optional[float] = None,
+    arg_buffer_meta=None,
+    arg_buffers=None,
 ):
     """One electron through a worker.  Returns (result, exception, meta)."""
     request = cloudpickle.dumps(
-        {"op_id": op_id, "workdir": workdir, "function_blob": function_blob}
+        {
+            "op_id": op_id,
+            "workdir": workdir,
+            "function_blob": function_blob,
+            "arg_buffers": arg_buffer_meta or [],
+        }
     )
     async with handle.channel.transaction() as ch:
         await ch.send_frame(request)
+        for view, _keep in arg_buffers or []:
+            await ch.send_frame(view)
         main = await ch.recv_frame(timeout=timeout)
         tag, result_blob, meta, nbuf = cloudpickle.loads(main)
         assert tag == "R1", f"unexpected worker response tag {tag!r}"

@@ -541,6 +541,8 @@ class SSHExecutor(RemoteExecutor):
         env: Optional[dict],
         gpu_lib: str,
         worker_tag: object,
+        arg_meta=None,
+        arg_bufs=None,
     ):
         """Run one electron on the persistent worker for ``worker_tag``
         (a GPU slot id, or a CPU worker index).  Respawns a dead worker
@@ -573,7 +575,12 @@ class SSHExecutor(RemoteExecutor):
                     continue
                 try:
                     return await worker_pool.run_task(
-                        handle, operation_id, workdir, function_blob
+                        handle,
+                        operation_id,
+                        workdir,
+                        function_blob,
+                        arg_buffer_meta=arg_meta,
+                        arg_buffers=arg_bufs,
                     )
                 except ChannelClosed:
                     worker_pool.drop(key)
@@ -728,7 +735,14 @@ class SSHExecutor(RemoteExecutor):
 
             if self.persistent_workers:
                 with timer.phase("stage"):
-                    function_blob = cloudpickle.dumps((function, args, kwargs))
+                    # large CPU-tensor arguments travel as raw frames, not
+                    # inside the pickle (mirror of the result staging)
+                    s_args, s_kwargs, arg_meta, arg_bufs = (
+                        worker_pool.extract_arg_buffers(
+                            args, kwargs, self.pinned_staging_threshold_bytes
+                        )
+                    )
+                    function_blob = cloudpickle.dumps((function, s_args, s_kwargs))
                 try:
                     with timer.phase("dispatch"):
                         result, exception, meta = await self._dispatch_worker(
@@ -738,6 +752,9 @@ class SSHExecutor(RemoteExecutor):
                             current_remote_workdir,
                             env,
                             gpu_lib,
+                            arg_meta=arg_meta,
+                            arg_bufs=arg_bufs,
+                            worker_tag=
                             (
                                 record.gpu_id
                                 if record.gpu_id is not None

@@ -30,11 +30,14 @@ class Channel:
     def alive(self) -> bool:
         return self._proc.returncode is None
 
-    async def send_frame(self, payload: bytes) -> None:
+    async def send_frame(self, payload) -> None:
+        """``payload`` is bytes or any buffer-protocol object (ctypes
+        views of pinned/tensor memory go out without copies)."""
         if not self.alive:
             raise ChannelClosed(f"{self._label}: process exited")
-        self._proc.stdin.write(struct.pack(">I", len(payload)))
-        self._proc.stdin.write(payload)
+        view = memoryview(payload).cast("B")
+        self._proc.stdin.write(struct.pack(">I", view.nbytes))
+        self._proc.stdin.write(view)
         await self._proc.stdin.drain()
 
     async def recv_frame(self, timeout: Optional[float] = None) -> bytes:

@@ -264,3 +264,36 @@ def test_worker_captures_task_output(local_executor):
     meta = ex.last_task_record.remote_meta
     assert "captured stdout line" in meta["stdout"]
     assert "captured stderr line" in meta["stderr"]
+
+
+def test_large_tensor_arguments_out_of_band(local_executor):
+    """Large CPU-tensor ARGUMENTS travel as raw request frames and
+    reconstruct exactly in the worker (mirror of result staging)."""
+    torch = __import__("pytest").importorskip("torch")
+    ex = local_executor(
+        persistent_workers=True, pinned_staging_threshold_bytes=1024
+    )
+
+    def consume(x, small, named=None):
+        import torch
+
+        assert isinstance(x, torch.Tensor)
+        return {
+            "sum": float(x.double().sum()),
+            "shape": tuple(x.shape),
+            "small_ok": bool((small == 1).all()),
+            "named_sum": float(named.double().sum()),
+        }
+
+    big = torch.arange(6000, dtype=torch.float32).reshape(3, -1)
+    small = torch.ones(4, dtype=torch.int64)  # below threshold: inline
+    named = torch.full((512,), 2.0)
+    out = asyncio.run(
+        ex.execute(consume, [big, small], {"named": named}, dispatch_id="ab", node_id=0)
+    )
+    assert out["sum"] == float(big.double().sum())
+    assert out["shape"] == (3, 2000)
+    assert out["small_ok"] is True
+    assert out["named_sum"] == 1024.0
+    # the original dispatcher-side tensors are untouched
+    assert torch.equal(big, torch.arange(6000, dtype=torch.float32).reshape(3, -1))
