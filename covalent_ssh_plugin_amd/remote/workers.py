@@ -69,7 +69,7 @@ def pick_cpu_tag(pool_key, pool_size: int) -> str:
     for tag in candidates:
         for key, handle in _workers.items():
             if len(key) >= 2 and key[0] == pool_key and key[1] == tag:
-                if handle.alive and not handle.channel._lock.locked():
+                if handle.alive and handle.channel.inflight == 0:
                     return tag
     return f"cpu{cpu_worker_index(pool_size)}"
 
@@ -205,14 +205,21 @@ async def run_task(
             "arg_buffers": arg_buffer_meta or [],
         }
     )
-    async with handle.channel.transaction() as ch:
-        await ch.send_frame(request)
-        for view, _keep in arg_buffers or []:
-            await ch.send_frame(view)
-        main = await ch.recv_frame(timeout=timeout)
+    frames = [request] + [view for view, _keep in (arg_buffers or [])]
+    parsed = {}
+
+    def reply_frames(main: bytes) -> int:
         tag, result_blob, meta, nbuf = cloudpickle.loads(main)
         assert tag == "R1", f"unexpected worker response tag {tag!r}"
-        buffers = [await ch.recv_frame(timeout=timeout) for _ in range(nbuf)]
+        parsed["value"] = (result_blob, meta, nbuf)
+        return nbuf
+
+    # pipelined: the channel's write lock is held only while sending, so
+    # queued electrons on one worker overlap their wire round trips
+    _main, buffers = await handle.channel.exchange(
+        frames, reply_frames, timeout=timeout
+    )
+    result_blob, meta, nbuf = parsed["value"]
     result, exception = cloudpickle.loads(result_blob)
     if nbuf:
         result = _reconstruct(result, meta.get("buffers", []), buffers)

@@ -23,12 +23,17 @@ class Channel:
     def __init__(self, proc: asyncio.subprocess.Process, label: str = "worker"):
         self._proc = proc
         self._label = label
-        self._lock = asyncio.Lock()  # one in-flight request per worker
+        self._lock = asyncio.Lock()  # guards the write side
+        self._inflight = 0  # requests sent, replies not yet resolved
         self.loop = asyncio.get_event_loop()
 
     @property
     def alive(self) -> bool:
         return self._proc.returncode is None
+
+    @property
+    def inflight(self) -> int:
+        return self._inflight
 
     async def send_frame(self, payload) -> None:
         """``payload`` is bytes or any buffer-protocol object (ctypes
@@ -81,9 +86,93 @@ class Channel:
         tensor-buffer frames)."""
         return _Transaction(self)
 
+    # -- pipelined exchange --------------------------------------------
+    #
+    # The worker protocol is strictly ordered: responses come back in
+    # request order.  ``exchange`` therefore only needs the write lock
+    # while SENDING its request frames; replies are consumed by a single
+    # reader pump that resolves waiter futures FIFO.  Over a real SSH
+    # link this removes the one-RTT-per-electron serialization the
+    # plain transaction() exchange pays.
+
+    def _ensure_pump(self) -> None:
+        if getattr(self, "_pump_task", None) is None or self._pump_task.done():
+            self._waiters: "asyncio.Queue" = asyncio.Queue()
+            self._pump_task = asyncio.get_running_loop().create_task(self._pump())
+
+    def _fail_waiters(self, first=None) -> None:
+        if first is not None and not first.done():
+            first.set_exception(ChannelClosed(f"{self._label}: channel down"))
+        waiters = getattr(self, "_waiters", None)
+        while waiters is not None and not waiters.empty():
+            w, _ = waiters.get_nowait()
+            if not w.done():
+                try:
+                    w.set_exception(ChannelClosed(f"{self._label}: channel down"))
+                except RuntimeError:
+                    pass  # waiter's loop already gone
+
+    async def _pump(self) -> None:
+        fut = None
+        try:
+            while True:
+                fut, reply_frames = await self._waiters.get()
+                main = await self.recv_frame()
+                extra_count = reply_frames(main)
+                extras = [await self.recv_frame() for _ in range(extra_count)]
+                if not fut.done():
+                    fut.set_result((main, extras))
+                fut = None
+        except asyncio.CancelledError:
+            # channel killed/closed: fail whatever is still waiting
+            self._fail_waiters(fut)
+            raise
+        except Exception as e:  # noqa: BLE001 - fail this + later waiters
+            if fut is not None and not fut.done():
+                fut.set_exception(
+                    e
+                    if isinstance(e, ChannelClosed)
+                    else ChannelClosed(f"{self._label}: reader failed: {e!r}")
+                )
+            self._fail_waiters()
+
+    async def exchange(
+        self,
+        request_frames,
+        reply_frames,
+        timeout: Optional[float] = None,
+    ):
+        """Pipelined request/response.
+
+        ``request_frames``: iterable of payloads sent back-to-back under
+        the write lock.  ``reply_frames``: callable(main_frame) -> number
+        of extra raw frames to read for this reply.  Returns
+        ``(main_frame, [extra_frames])``.  Multiple callers may have
+        requests in flight concurrently; replies resolve in order.
+        """
+        self._ensure_pump()
+        fut = asyncio.get_running_loop().create_future()
+        self._inflight += 1
+        try:
+            async with self._lock:
+                self._waiters.put_nowait((fut, reply_frames))
+                for frame in request_frames:
+                    await self.send_frame(frame)
+            if timeout is None:
+                return await fut
+            return await asyncio.wait_for(fut, timeout=timeout)
+        finally:
+            self._inflight -= 1
+
     def kill(self) -> None:
         """Synchronous hard-kill (for reaping workers whose event loop is
         already gone)."""
+        task = getattr(self, "_pump_task", None)
+        if task is not None and not task.done():
+            try:
+                task.cancel()
+            except RuntimeError:
+                pass  # task's loop already closed
         if self.alive:
             try:
                 self._proc.kill()
@@ -91,6 +180,9 @@ class Channel:
                 pass
 
     async def close(self) -> None:
+        task = getattr(self, "_pump_task", None)
+        if task is not None and not task.done():
+            task.cancel()
         if self.loop is not asyncio.get_event_loop() or self.loop.is_closed():
             # channel belongs to another (likely closed) loop: its pipe
             # transports cannot be driven from here — hard-kill instead

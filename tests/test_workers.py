@@ -297,3 +297,39 @@ def test_large_tensor_arguments_out_of_band(local_executor):
     assert out["named_sum"] == 1024.0
     # the original dispatcher-side tensors are untouched
     assert torch.equal(big, torch.arange(6000, dtype=torch.float32).reshape(3, -1))
+
+
+def test_worker_requests_pipeline_on_one_channel(local_executor):
+    """Multiple queued electrons on ONE worker are all written to the
+    channel without waiting for earlier replies (pipelining): the
+    channel's in-flight count exceeds 1 while the first task runs."""
+    from covalent_ssh_plugin_amd.remote import workers as worker_pool
+
+    ex = local_executor(persistent_workers=True, cpu_workers=1)
+
+    def slow(i):
+        import time
+
+        time.sleep(0.3)
+        return i
+
+    async def main():
+        tasks = [
+            asyncio.ensure_future(
+                ex.execute(slow, [i], {}, dispatch_id="pl", node_id=i)
+            )
+            for i in range(3)
+        ]
+        peak = 0
+        for _ in range(300):
+            for handle in worker_pool._workers.values():
+                peak = max(peak, handle.channel.inflight)
+            if all(t.done() for t in tasks):
+                break
+            await asyncio.sleep(0.01)
+        results = await asyncio.gather(*tasks)
+        return results, peak
+
+    results, peak = asyncio.run(main())
+    assert results == [0, 1, 2]
+    assert peak >= 2, f"requests were not pipelined (peak in-flight {peak})"
