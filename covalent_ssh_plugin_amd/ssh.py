@@ -203,6 +203,14 @@ class SSHExecutor(RemoteExecutor):
         self.last_task_record: Optional[TaskRecord] = None
         #: bounded history of task records (bench reads this)
         self.task_records: List[TaskRecord] = []
+        #: operational counters (observability)
+        self.counters: Dict[str, int] = {
+            "tasks": 0,
+            "task_exceptions": 0,
+            "worker_respawns": 0,
+            "cancellations": 0,
+            "ssh_failures": 0,
+        }
 
     # ------------------------------------------------------------------
     # Endpoint identity / pooled state
@@ -237,6 +245,7 @@ class SSHExecutor(RemoteExecutor):
     ) -> Any:
         """If ``run_local_on_ssh_fail``, run the task on the dispatcher
         host; otherwise raise (reference ssh.py:202-208)."""
+        self.counters["ssh_failures"] += 1
         app_log.warning("SSH dispatch failed: %s", message)
         if self.run_local_on_ssh_fail:
             return await asyncio.to_thread(function, *args, **kwargs)
@@ -365,6 +374,7 @@ class SSHExecutor(RemoteExecutor):
                 f"persistent workers (no in-flight worker task {operation_id!r})"
             )
         self._cancelled.add(operation_id)
+        self.counters["cancellations"] += 1
         await worker_pool.kill(key)
 
     # ------------------------------------------------------------------
@@ -589,6 +599,7 @@ class SSHExecutor(RemoteExecutor):
                         raise SSHTaskError(f"task {operation_id} was cancelled")
                     if attempt == 1:
                         raise
+                    self.counters["worker_respawns"] += 1
                     app_log.warning(
                         "worker %s died mid-task; respawning once", key
                     )
@@ -872,6 +883,9 @@ class SSHExecutor(RemoteExecutor):
 
         record.phases = timer.snapshot()
         record.total_s = timer.total()
+        self.counters["tasks"] += 1
+        if exception is not None:
+            self.counters["task_exceptions"] += 1
         self.last_task_record = record
         self.task_records.append(record)
         log_path = os.environ.get("CSP_AMD_TASK_LOG")
@@ -940,7 +954,9 @@ class SSHExecutor(RemoteExecutor):
         completed tasks (observability; SURVEY.md §5 metrics row)."""
         from .utils.timing import summarize
 
-        return summarize(self.task_records)
+        out = summarize(self.task_records)
+        out["counters"] = dict(self.counters)
+        return out
 
     # Convenience for closing pooled transports (e.g. at interpreter exit
     # or between tests).  The reference closes per-task; pooled transports

@@ -229,3 +229,23 @@ def test_cold_start_env_checks_run_once(local_executor):
     out = asyncio.run(main())
     assert out == [i + 1 for i in range(8)]
     assert len(calls) == 1, calls
+
+
+def test_counters(local_executor):
+    ex = local_executor()
+
+    def boom():
+        raise ValueError("x")
+
+    async def main():
+        await ex.execute(_add, [1, 2], {}, dispatch_id="cn", node_id=0)
+        try:
+            await ex.execute(boom, [], {}, dispatch_id="cn", node_id=1)
+        except ValueError:
+            pass
+
+    asyncio.run(main())
+    s = ex.stats()
+    assert s["counters"]["tasks"] == 2
+    assert s["counters"]["task_exceptions"] == 1
+    assert s["counters"]["ssh_failures"] == 0
