@@ -83,6 +83,7 @@ _EXECUTOR_PLUGIN_DEFAULTS = {
     "batch_roundtrips": True,  # fused single-round-trip dispatch
     "persistent_workers": False,  # warm worker process per GPU slot
     "cpu_workers": 4,  # worker-set size when no GPU policy is active
+    "task_timeout": 0,  # seconds; 0 = unlimited
 }
 
 update_config_defaults("executors.ssh", _EXECUTOR_PLUGIN_DEFAULTS)
@@ -148,6 +149,7 @@ class SSHExecutor(RemoteExecutor):
         batch_roundtrips: Optional[bool] = None,
         persistent_workers: Optional[bool] = None,
         cpu_workers: Optional[int] = None,
+        task_timeout: Optional[float] = None,
         fixed_gpu: int = 0,
         local_home: str = "",
     ) -> None:
@@ -192,6 +194,7 @@ class SSHExecutor(RemoteExecutor):
             _conf("persistent_workers", persistent_workers, default=False)
         )
         self.cpu_workers = int(_conf("cpu_workers", cpu_workers))
+        self.task_timeout = float(_conf("task_timeout", task_timeout, default=0) or 0)
         self.fixed_gpu = int(fixed_gpu)
         self.local_home = local_home
 
@@ -589,8 +592,17 @@ class SSHExecutor(RemoteExecutor):
                         operation_id,
                         workdir,
                         function_blob,
+                        timeout=self.task_timeout or None,
                         arg_buffer_meta=arg_meta,
                         arg_buffers=arg_bufs,
+                    )
+                except asyncio.TimeoutError:
+                    # the worker is wedged on this task: kill it so the
+                    # slot/worker is usable again, then surface
+                    await worker_pool.kill(key)
+                    raise SSHTaskError(
+                        f"task {operation_id} exceeded task_timeout="
+                        f"{self.task_timeout}s"
                     )
                 except ChannelClosed:
                     worker_pool.drop(key)
@@ -663,7 +675,9 @@ class SSHExecutor(RemoteExecutor):
             f"printf '%s' {q(s_meta)}; cat {q(paths['meta_remote'])} 2>/dev/null || true; "
             f"fi; {cleanup}exit $_csp_rc"
         )
-        proc = await transport.run(cmd, input_data=tar_bytes, env=env)
+        proc = await transport.run(
+            cmd, input_data=tar_bytes, env=env, timeout=self.task_timeout or None
+        )
         task_out, result_bytes, meta_bytes = self._split_stream(
             proc.stdout, s_result.encode(), s_meta.encode()
         )
@@ -791,6 +805,12 @@ class SSHExecutor(RemoteExecutor):
                         proc, result_bytes, meta_bytes = await self._dispatch_fused(
                             transport, paths, env
                         )
+                except asyncio.TimeoutError:
+                    self._cleanup_local(paths)
+                    raise SSHTaskError(
+                        f"task {operation_id} exceeded task_timeout="
+                        f"{self.task_timeout}s"
+                    )
                 except (TransportConnectError, OSError) as e:
                     self._cleanup_local(paths)
                     return await self._on_ssh_fail(

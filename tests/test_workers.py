@@ -333,3 +333,41 @@ def test_worker_requests_pipeline_on_one_channel(local_executor):
     results, peak = asyncio.run(main())
     assert results == [0, 1, 2]
     assert peak >= 2, f"requests were not pipelined (peak in-flight {peak})"
+
+
+def test_task_timeout_worker_mode(local_executor):
+    """A task exceeding task_timeout fails promptly; the wedged worker is
+    killed and replaced for the next electron."""
+    ex = local_executor(persistent_workers=True, cpu_workers=1, task_timeout=1.0)
+
+    def hang():
+        import time
+
+        time.sleep(60)
+
+    def quick():
+        return "ok"
+
+    import time
+
+    t0 = time.perf_counter()
+    with pytest.raises(RuntimeError, match="task_timeout"):
+        asyncio.run(ex.execute(hang, [], {}, dispatch_id="to", node_id=0))
+    assert time.perf_counter() - t0 < 10
+    assert asyncio.run(ex.execute(quick, [], {}, dispatch_id="to", node_id=1)) == "ok"
+
+
+def test_task_timeout_fused_mode(local_executor):
+    ex = local_executor(task_timeout=1.0)
+
+    def hang():
+        import time
+
+        time.sleep(60)
+
+    import time
+
+    t0 = time.perf_counter()
+    with pytest.raises(RuntimeError, match="task_timeout"):
+        asyncio.run(ex.execute(hang, [], {}, dispatch_id="tf", node_id=0))
+    assert time.perf_counter() - t0 < 10
