@@ -6,8 +6,14 @@ Public surface mirrors the reference package
 ``covalent.executor.executor_plugins`` entry point (setup.py).
 """
 
+from .cluster import SSHClusterExecutor
 from .ssh import _EXECUTOR_PLUGIN_DEFAULTS, EXECUTOR_PLUGIN_NAME, SSHExecutor
 
-__all__ = ["SSHExecutor", "EXECUTOR_PLUGIN_NAME", "_EXECUTOR_PLUGIN_DEFAULTS"]
+__all__ = [
+    "SSHExecutor",
+    "SSHClusterExecutor",
+    "EXECUTOR_PLUGIN_NAME",
+    "_EXECUTOR_PLUGIN_DEFAULTS",
+]
 
 __version__ = "0.1.0"
