@@ -308,13 +308,9 @@ class SSHExecutor(RemoteExecutor):
         """Remote command string: ``{python_path} {script}`` with the
         conda activation wrapper when ``conda_env`` is set (reference
         ssh.py:377-380)."""
-        cmd = f"{self.python_path} {shlex.quote(remote_script_file)}"
-        if self.conda_env:
-            cmd = (
-                'eval "$(conda shell.bash hook)" && '
-                f"conda activate {shlex.quote(self.conda_env)} && {cmd}"
-            )
-        return cmd
+        return self._wrap_conda(
+            f"{self.python_path} {shlex.quote(remote_script_file)}"
+        )
 
     async def submit_task(
         self,
