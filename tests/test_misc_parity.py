@@ -73,3 +73,26 @@ def test_executor_instance_is_picklable():
     assert ex2.hostname == "h"
     assert ex2.gpu_slots == 4
     assert ex2.persistent_workers is True
+
+
+def test_pickled_executor_still_executes(tmp_path):
+    """Covalent deserializes executor instances before calling run();
+    a round-tripped instance must dispatch normally (pooled state is
+    module-level and re-binds on first use)."""
+    import asyncio
+    import pickle
+
+    home = tmp_path / "home"
+    home.mkdir()
+    ex = SSHExecutor(
+        transport="local",
+        local_home=str(home),
+        cache_dir=str(tmp_path / "cache"),
+        python_path=sys.executable,
+    )
+    ex2 = pickle.loads(pickle.dumps(ex))
+
+    def fn(a, b):
+        return a * b
+
+    assert asyncio.run(ex2.execute(fn, [6, 7], {})) == 42
