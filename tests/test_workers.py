@@ -371,3 +371,21 @@ def test_task_timeout_fused_mode(local_executor):
     with pytest.raises(RuntimeError, match="task_timeout"):
         asyncio.run(ex.execute(hang, [], {}, dispatch_id="tf", node_id=0))
     assert time.perf_counter() - t0 < 10
+
+
+def test_unpicklable_result_surfaces_as_error(local_executor):
+    """A result cloudpickle cannot serialize (e.g. a generator) comes
+    back as a pickling exception, not a dead worker."""
+    ex = local_executor(persistent_workers=True)
+
+    def gen():
+        return (i for i in range(3))
+
+    with pytest.raises(Exception) as exc_info:
+        asyncio.run(ex.execute(gen, [], {}, dispatch_id="up", node_id=0))
+    assert "pickle" in str(exc_info.value).lower() or "generator" in str(
+        exc_info.value
+    ).lower(), exc_info.value
+
+    # worker survives and serves the next electron
+    assert asyncio.run(ex.execute(lambda: 3, [], {}, dispatch_id="up", node_id=1)) == 3
