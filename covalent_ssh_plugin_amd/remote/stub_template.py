@@ -262,7 +262,10 @@ def main():
             result, exception = None, e
         _mark("staging", t_stage)
 
-    blob = pickle.dumps((result, exception))
+    try:
+        blob = pickle.dumps((result, exception))
+    except Exception as e:  # noqa: BLE001 - unpicklable result
+        blob = pickle.dumps((None, e))
     with open(RESULT_FILE, "wb") as f_out:
         f_out.write(blob)
     # integrity check for the sentinel-framed fused return path

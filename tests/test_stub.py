@@ -155,3 +155,16 @@ def test_gpu_slot_resolves_within_ambient_visibility(tmp_path):
     proc, result_file, _ = _run_stub(tmp_path, visible, env=env)
     result, _ = pickle.loads(result_file.read_bytes())
     assert result == "3"
+
+
+def test_unpicklable_result_in_pickle_not_crash(tmp_path):
+    """An unpicklable result becomes (None, TypeError) in the result
+    pickle with exit 0 (the worker path behaves the same)."""
+
+    def gen():
+        return (i for i in range(3))
+
+    proc, result_file, _ = _run_stub(tmp_path, gen)
+    assert proc.returncode == 0, proc.stderr
+    result, exception = pickle.loads(result_file.read_bytes())
+    assert result is None and isinstance(exception, TypeError)
