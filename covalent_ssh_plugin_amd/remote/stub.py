@@ -59,6 +59,7 @@ def render_worker(
     gpu_lib_path: str = "",
     warmup: bool = True,
     staging_threshold: Optional[int] = None,
+    idle_timeout: float = 0.0,
 ) -> str:
     """Return the persistent worker script text (one per endpoint; the
     GPU slot arrives via the CSP_GPU_SLOT env var at launch)."""
@@ -68,6 +69,7 @@ def render_worker(
         "__CSP_GPU_LIB__": gpu_lib_path,
         "__CSP_WARMUP__": repr(bool(warmup)),
         "__CSP_STAGING_THRESHOLD__": str(thr),
+        "__CSP_IDLE_TIMEOUT__": repr(float(idle_timeout)),
     }
     for token, value in replacements.items():
         text = text.replace(token, value)

@@ -44,6 +44,7 @@ import time
 GPU_LIB = "__CSP_GPU_LIB__"
 DO_WARMUP = bool(__CSP_WARMUP__)
 STAGING_THRESHOLD = int(__CSP_STAGING_THRESHOLD__)
+IDLE_TIMEOUT = float(__CSP_IDLE_TIMEOUT__)  # seconds; 0 = never exit
 
 if GPU_LIB:
     GPU_LIB = os.path.abspath(os.path.expanduser(GPU_LIB))
@@ -225,7 +226,13 @@ def _stage_result(result, stats, buffers, buffer_meta):
     return walk(result)
 
 
-def _read_frame(fd):
+def _read_frame(fd, idle_timeout=0.0):
+    if idle_timeout > 0:
+        import select
+
+        ready, _, _ = select.select([fd], [], [], idle_timeout)
+        if not ready:
+            return None  # idle too long: orderly exit
     header = b""
     while len(header) < 4:
         chunk = os.read(fd, 4 - len(header))
@@ -364,7 +371,7 @@ def main():
         sys.exit(3)
 
     while True:
-        frame = _read_frame(0)
+        frame = _read_frame(0, idle_timeout=IDLE_TIMEOUT)
         if frame is None or frame == b"":
             break
         request = pickle.loads(frame)

@@ -84,6 +84,7 @@ _EXECUTOR_PLUGIN_DEFAULTS = {
     "persistent_workers": False,  # warm worker process per GPU slot
     "cpu_workers": 4,  # worker-set size when no GPU policy is active
     "task_timeout": 0,  # seconds; 0 = unlimited
+    "worker_idle_timeout": 0,  # seconds; 0 = workers never exit on idle
 }
 
 update_config_defaults("executors.ssh", _EXECUTOR_PLUGIN_DEFAULTS)
@@ -150,6 +151,7 @@ class SSHExecutor(RemoteExecutor):
         persistent_workers: Optional[bool] = None,
         cpu_workers: Optional[int] = None,
         task_timeout: Optional[float] = None,
+        worker_idle_timeout: Optional[float] = None,
         fixed_gpu: int = 0,
         local_home: str = "",
     ) -> None:
@@ -195,6 +197,9 @@ class SSHExecutor(RemoteExecutor):
         )
         self.cpu_workers = int(_conf("cpu_workers", cpu_workers))
         self.task_timeout = float(_conf("task_timeout", task_timeout, default=0) or 0)
+        self.worker_idle_timeout = float(
+            _conf("worker_idle_timeout", worker_idle_timeout, default=0) or 0
+        )
         self.fixed_gpu = int(fixed_gpu)
         self.local_home = local_home
 
@@ -514,6 +519,7 @@ class SSHExecutor(RemoteExecutor):
             gpu_lib_path=gpu_lib,
             warmup=self.warmup_gpu,
             staging_threshold=self.pinned_staging_threshold_bytes,
+            idle_timeout=self.worker_idle_timeout,
         )
         digest = _script_digest(text)
         key = self._pool_key()

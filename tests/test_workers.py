@@ -389,3 +389,34 @@ def test_unpicklable_result_surfaces_as_error(local_executor):
 
     # worker survives and serves the next electron
     assert asyncio.run(ex.execute(lambda: 3, [], {}, dispatch_id="up", node_id=1)) == 3
+
+
+def test_worker_idle_timeout_exit_and_respawn(local_executor):
+    """With worker_idle_timeout set, an idle worker exits on its own; the
+    next electron transparently respawns one."""
+    import time
+
+    ex = local_executor(
+        persistent_workers=True, cpu_workers=1, worker_idle_timeout=0.5
+    )
+
+    def pidof():
+        import os
+
+        return os.getpid()
+
+    async def main():
+        pid1 = await ex.execute(pidof, [], {}, dispatch_id="it", node_id=0)
+        from covalent_ssh_plugin_amd.remote import workers as worker_pool
+
+        (handle,) = worker_pool._workers.values()
+        for _ in range(100):  # wait for the idle exit
+            if not handle.alive:
+                break
+            await asyncio.sleep(0.1)
+        assert not handle.alive, "worker did not exit on idle timeout"
+        pid2 = await ex.execute(pidof, [], {}, dispatch_id="it", node_id=1)
+        return pid1, pid2
+
+    pid1, pid2 = asyncio.run(main())
+    assert pid1 != pid2
