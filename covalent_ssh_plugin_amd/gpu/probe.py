@@ -62,6 +62,8 @@ def load(path: str = "") -> ctypes.CDLL:
     lib.csp_device_count.restype = ctypes.c_int
     lib.csp_probe_json.restype = ctypes.c_int
     lib.csp_probe_json.argtypes = [ctypes.c_int, ctypes.c_char_p, ctypes.c_size_t]
+    lib.csp_probe_props_json.restype = ctypes.c_int
+    lib.csp_probe_props_json.argtypes = [ctypes.c_int, ctypes.c_char_p, ctypes.c_size_t]
     lib.csp_warmup.restype = ctypes.c_int
     lib.csp_warmup.argtypes = [ctypes.c_int, ctypes.c_int]
     lib.csp_host_alloc.restype = ctypes.c_void_p
@@ -105,6 +107,14 @@ def probe(device: int = 0) -> dict:
     lib = load()
     buf = ctypes.create_string_buffer(8192)
     _check(lib, lib.csp_probe_json(device, buf, len(buf)), "csp_probe_json")
+    return json.loads(buf.value.decode())
+
+
+def probe_props(device: int = 0) -> dict:
+    """Device properties only (no measurement kernels, no big allocs)."""
+    lib = load()
+    buf = ctypes.create_string_buffer(8192)
+    _check(lib, lib.csp_probe_props_json(device, buf, len(buf)), "csp_probe_props_json")
     return json.loads(buf.value.decode())
 
 

@@ -218,6 +218,31 @@ extern "C" int csp_warmup(int device, int budget_ms) {
     return rc;
 }
 
+// Properties-only probe: no measurement kernels, no large allocations —
+// cheap enough for per-task use (the stub's prologue).  Measured
+// bandwidth/TFLOPs fields are 0; csp_probe_json fills them.
+extern "C" int csp_probe_props_json(int device, char* buf, size_t buflen) {
+    HIP_TRY(hipSetDevice(device));
+    hipDeviceProp_t props;
+    HIP_TRY(hipGetDeviceProperties(&props, device));
+    size_t mem_free = 0, mem_total = 0;
+    HIP_TRY(hipMemGetInfo(&mem_free, &mem_total));
+    int written = snprintf(
+        buf, buflen,
+        "{\"name\":\"%s\",\"gcn_arch\":\"%s\",\"device\":%d,"
+        "\"cu_count\":%d,\"max_clock_mhz\":%d,\"lds_per_cu_kb\":%zu,"
+        "\"wavefront_size\":%d,\"hbm_total_gb\":%.1f,\"hbm_free_gb\":%.1f,"
+        "\"hbm_bw_gbps\":0,\"mfma_bf16_tflops\":0}",
+        props.name, props.gcnArchName, device, props.multiProcessorCount,
+        props.clockRate / 1000, (size_t)props.maxSharedMemoryPerMultiProcessor / 1024,
+        props.warpSize, (double)mem_total / 1.0e9, (double)mem_free / 1.0e9);
+    if (written < 0 || (size_t)written >= buflen) {
+        snprintf(g_err, sizeof(g_err), "probe json buffer too small");
+        return -2;
+    }
+    return 0;
+}
+
 extern "C" int csp_probe_json(int device, char* buf, size_t buflen) {
     HIP_TRY(hipSetDevice(device));
     hipDeviceProp_t props;

@@ -122,8 +122,8 @@ def _load_gpu_lib():
     _ensure_torch_runtime_first()
 
     lib = ctypes.CDLL(GPU_LIB)
-    lib.csp_probe_json.restype = ctypes.c_int
-    lib.csp_probe_json.argtypes = [ctypes.c_int, ctypes.c_char_p, ctypes.c_size_t]
+    lib.csp_probe_props_json.restype = ctypes.c_int
+    lib.csp_probe_props_json.argtypes = [ctypes.c_int, ctypes.c_char_p, ctypes.c_size_t]
     lib.csp_warmup.restype = ctypes.c_int
     lib.csp_warmup.argtypes = [ctypes.c_int, ctypes.c_int]
     lib.csp_staging_get.restype = ctypes.c_void_p
@@ -140,11 +140,15 @@ def _gpu_prologue():
 
     t = time.monotonic()
     lib = _load_gpu_lib()
+    # per-task prologue: cheap props probe (no measurement kernels) +
+    # short warm-up spin; the persistent worker does the full measured
+    # probe once instead
     buf = ctypes.create_string_buffer(8192)
-    rc = lib.csp_probe_json(0, buf, len(buf))
+    rc = lib.csp_probe_props_json(0, buf, len(buf))
     if rc != 0:
         raise RuntimeError(
-            "csp_probe_json failed: %s" % lib.csp_last_error().decode(errors="replace")
+            "csp_probe_props_json failed: %s"
+            % lib.csp_last_error().decode(errors="replace")
         )
     _meta["gpu"] = json.loads(buf.value.decode())
     if DO_WARMUP:

@@ -212,3 +212,16 @@ def test_rccl_allreduce_smoke_gpu(tmp_path):
     assert out["correct"] is True
     assert out["world_size"] == world
     print("rccl busbw GB/s:", out["busbw_GBps"])
+
+
+def test_probe_props_fast(gpu_lib):
+    import time
+
+    from covalent_ssh_plugin_amd.gpu import probe
+
+    t0 = time.perf_counter()
+    info = probe.probe_props(0)
+    dt = time.perf_counter() - t0
+    assert "gfx950" in info["gcn_arch"]
+    assert info["cu_count"] == 256
+    assert dt < 1.0, dt  # no measurement kernels in the props probe
