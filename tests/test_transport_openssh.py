@@ -167,3 +167,32 @@ def test_real_ssh_client_connect_refused(tmp_path):
     with pytest.raises(TransportConnectError):
         asyncio.run(t.connect())
     assert not t.is_connected
+
+
+def test_open_channel_argv(tmp_path, monkeypatch):
+    """Worker channels ride the same multiplexed ssh argv, with the slot
+    env exported in the remote command."""
+    captured = {}
+
+    async def fake_exec(*argv, **kw):
+        captured["argv"] = argv
+
+        class P:
+            returncode = None
+            stdin = stdout = None
+
+        return P()
+
+    monkeypatch.setattr(asyncio, "create_subprocess_exec", fake_exec)
+    t = make_transport(tmp_path)
+
+    async def main():
+        await t.open_channel("python worker.py", env={"CSP_GPU_SLOT": "5"})
+
+    asyncio.run(main())
+    argv = captured["argv"]
+    assert argv[0] == "ssh"
+    assert "node0" in argv
+    remote_cmd = argv[-1]
+    assert remote_cmd == "export CSP_GPU_SLOT=5 && python worker.py"
+    assert f"ControlPath={t._control_path}" in " ".join(argv)
