@@ -225,3 +225,24 @@ def test_probe_props_fast(gpu_lib):
     assert "gfx950" in info["gcn_arch"]
     assert info["cu_count"] == 256
     assert dt < 1.0, dt  # no measurement kernels in the props probe
+
+
+def test_pinned_h2d_roundtrip(gpu_lib):
+    """csp_memcpy_h2d: pinned host -> device -> back is bit-exact."""
+    import ctypes
+
+    nbytes = 1 << 20
+    src = torch.randint(0, 255, (nbytes,), dtype=torch.uint8)
+    host = gpu_lib.csp_host_alloc(nbytes)
+    assert host
+    try:
+        ctypes.memmove(host, src.data_ptr(), nbytes)
+        dev = torch.empty(nbytes, dtype=torch.uint8, device="cuda")
+        torch.cuda.synchronize()
+        rc = gpu_lib.csp_memcpy_h2d(
+            ctypes.c_void_p(dev.data_ptr()), ctypes.c_void_p(host), nbytes
+        )
+        assert rc == 0
+        assert torch.equal(dev.cpu(), src)
+    finally:
+        gpu_lib.csp_host_free(ctypes.c_void_p(host))
