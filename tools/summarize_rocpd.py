@@ -60,6 +60,28 @@ def summarize(path: str) -> None:
         )
     print()
 
+    # memory copies (present with --memory-copy-trace / sys-trace runs)
+    try:
+        mc = table(con, "rocpd_memory_copy")
+        mc_rows = con.execute(
+            f"""SELECT name_id, COUNT(*), SUM(size)/1e9,
+                       SUM(end-start)/1e6 FROM {mc} GROUP BY name_id"""
+        ).fetchall()
+        if mc_rows:
+            strt = table(con, "rocpd_string")
+            print("| memory copy | count | GB | ms | GB/s |")
+            print("|---|---|---|---|---|")
+            for name_id, cnt, gb, ms in mc_rows:
+                row = con.execute(
+                    f"SELECT string FROM {strt} WHERE id=?", (name_id,)
+                ).fetchone()
+                name = row[0] if row else str(name_id)
+                rate = gb / (ms / 1e3) if ms else 0.0
+                print(f"| {name} | {cnt} | {gb:.2f} | {ms:.1f} | {rate:.1f} |")
+            print()
+    except SystemExit:
+        pass
+
 
 if __name__ == "__main__":
     for p in sys.argv[1:]:
