@@ -8,7 +8,8 @@ python + HIP-runtime start per electron (the reference's architecture,
 /root/reference/covalent_ssh_plugin/exec.py), a worker pays it once:
 
   * starts, optionally runs the CDNA4 warm-up/device-probe prologue on
-    its pinned GPU (HIP_VISIBLE_DEVICES is set by the launcher),
+    its pinned GPU (the launcher injects CSP_GPU_SLOT; _resolve_gpu_slot
+    maps it into the ambient HIP_VISIBLE_DEVICES),
   * then serves electrons over a length-framed binary protocol on
     stdin/stdout: each request carries the cloudpickled
     ``(fn, args, kwargs)`` and a workdir; each reply carries the pickled
