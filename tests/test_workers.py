@@ -420,3 +420,53 @@ def test_worker_idle_timeout_exit_and_respawn(local_executor):
 
     pid1, pid2 = asyncio.run(main())
     assert pid1 != pid2
+
+
+def test_no_orphan_workers_when_dispatcher_dies(tmp_path):
+    """A dispatcher that exits WITHOUT close_pool must not leave worker
+    processes behind: the worker sees EOF on its stdin and exits."""
+    import subprocess
+    import sys
+    import time
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parent.parent
+    code = f"""
+import asyncio, os, sys, tempfile
+sys.path.insert(0, {str(repo)!r})
+from covalent_ssh_plugin_amd import SSHExecutor
+
+async def main():
+    home = tempfile.mkdtemp()
+    cache = tempfile.mkdtemp()
+    ex = SSHExecutor(transport="local", local_home=home, cache_dir=cache,
+                     python_path=sys.executable, persistent_workers=True,
+                     cpu_workers=1)
+    def pidof():
+        import os
+        return os.getpid()
+    pid = await ex.execute(pidof, [], {{}})
+    print(pid, flush=True)
+    os._exit(0)  # die abruptly: no close_pool, no cleanup
+
+asyncio.run(main())
+"""
+    proc = subprocess.run(
+        [sys.executable, "-c", code], capture_output=True, text=True, timeout=120
+    )
+    assert proc.returncode == 0, proc.stderr[-1500:]
+    worker_pid = int(proc.stdout.strip().splitlines()[-1])
+    # the worker must exit on its own (EOF on stdin) within a few seconds
+    for _ in range(50):
+        try:
+            import os
+
+            os.kill(worker_pid, 0)  # still alive?
+        except ProcessLookupError:
+            break
+        time.sleep(0.1)
+    else:
+        import os
+
+        os.kill(worker_pid, 9)  # clean up before failing
+        raise AssertionError(f"worker {worker_pid} orphaned after dispatcher death")
