@@ -470,3 +470,73 @@ asyncio.run(main())
 
         os.kill(worker_pid, 9)  # clean up before failing
         raise AssertionError(f"worker {worker_pid} orphaned after dispatcher death")
+
+
+def test_worker_standalone_protocol_and_slot_resolution(tmp_path):
+    """Drive a rendered worker directly over its framed protocol (no
+    executor): READY handshake, slot resolution within ambient
+    HIP_VISIBLE_DEVICES, task round trip, orderly shutdown."""
+    import os
+    import struct
+    import subprocess
+    import sys
+
+    import cloudpickle
+
+    from covalent_ssh_plugin_amd.remote.stub import render_worker
+
+    script = tmp_path / "worker.py"
+    script.write_text(render_worker())
+    env = dict(os.environ)
+    env["CSP_GPU_SLOT"] = "1"
+    env["HIP_VISIBLE_DEVICES"] = "4,6"
+    proc = subprocess.Popen(
+        [sys.executable, str(script)],
+        stdin=subprocess.PIPE,
+        stdout=subprocess.PIPE,
+        stderr=subprocess.DEVNULL,
+        env=env,
+        cwd=tmp_path,
+    )
+
+    def send(payload: bytes):
+        proc.stdin.write(struct.pack(">I", len(payload)) + payload)
+        proc.stdin.flush()
+
+    def recv() -> bytes:
+        header = proc.stdout.read(4)
+        (n,) = struct.unpack(">I", header)
+        return proc.stdout.read(n)
+
+    try:
+        tag, startup = cloudpickle.loads(recv())
+        assert tag == "READY"
+        assert startup["gpu_slot"] == "1"
+        assert startup["hip_visible_devices"] == "6"  # ambient[1]
+
+        def fn():
+            import os
+
+            return os.environ["HIP_VISIBLE_DEVICES"]
+
+        send(
+            cloudpickle.dumps(
+                {
+                    "op_id": "x_0",
+                    "workdir": str(tmp_path / "wd"),
+                    "function_blob": cloudpickle.dumps((fn, [], {})),
+                    "arg_buffers": [],
+                }
+            )
+        )
+        tag, blob, meta, nbuf = cloudpickle.loads(recv())
+        assert tag == "R1" and nbuf == 0
+        result, exception = cloudpickle.loads(blob)
+        assert exception is None
+        assert result == "6"
+
+        send(b"")  # zero-length frame: orderly shutdown
+        assert proc.wait(timeout=10) == 0
+    finally:
+        if proc.poll() is None:
+            proc.kill()
