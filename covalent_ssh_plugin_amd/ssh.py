@@ -153,6 +153,7 @@ class SSHExecutor(RemoteExecutor):
         task_timeout: Optional[float] = None,
         worker_idle_timeout: Optional[float] = None,
         fixed_gpu: int = 0,
+        ssh_extra_options: Optional[List[str]] = None,
         local_home: str = "",
     ) -> None:
         remote_cache = _conf("remote_cache", remote_cache)
@@ -201,6 +202,10 @@ class SSHExecutor(RemoteExecutor):
             _conf("worker_idle_timeout", worker_idle_timeout, default=0) or 0
         )
         self.fixed_gpu = int(fixed_gpu)
+        #: raw `ssh -o`/flag passthrough, e.g. ["-o", "StrictHostKeyChecking=yes",
+        #: "-J", "bastion"] (the default host-key policy matches the
+        #: reference: known_hosts disabled, reference ssh.py:267)
+        self.ssh_extra_options = list(ssh_extra_options or [])
         self.local_home = local_home
 
         #: in-flight worker-dispatched tasks: operation_id -> worker key
@@ -227,7 +232,14 @@ class SSHExecutor(RemoteExecutor):
     def _pool_key(self) -> Tuple[str, ...]:
         if self.transport_kind == "local":
             return ("local", self.local_home or os.path.expanduser("~"))
-        return ("ssh", self.hostname, self.username, self.ssh_key_file, str(self.ssh_port))
+        return (
+            "ssh",
+            self.hostname,
+            self.username,
+            self.ssh_key_file,
+            str(self.ssh_port),
+            ",".join(self.ssh_extra_options),
+        )
 
     def _make_transport(self) -> Transport:
         if self.transport_kind == "local":
@@ -237,6 +249,7 @@ class SSHExecutor(RemoteExecutor):
             username=self.username,
             ssh_key_file=self.ssh_key_file,
             port=self.ssh_port,
+            extra_options=self.ssh_extra_options,
         )
 
     def _slot_table(self) -> gpu_slots.SlotTable:

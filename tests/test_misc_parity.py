@@ -96,3 +96,18 @@ def test_pickled_executor_still_executes(tmp_path):
         return a * b
 
     assert asyncio.run(ex2.execute(fn, [6, 7], {})) == 42
+
+
+def test_ssh_extra_options_passthrough():
+    ex = SSHExecutor(
+        username="u",
+        hostname="h",
+        ssh_extra_options=["-o", "StrictHostKeyChecking=yes", "-J", "bastion"],
+    )
+    t = ex._make_transport()
+    args = t._base_args()
+    assert "StrictHostKeyChecking=yes" in args
+    assert "-J" in args and "bastion" in args
+    # different option sets must not share a pooled transport
+    ex2 = SSHExecutor(username="u", hostname="h")
+    assert ex._pool_key() != ex2._pool_key()
