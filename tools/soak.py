@@ -51,9 +51,34 @@ def main() -> None:
     ap.add_argument("--tensor-bytes", type=int, default=0,
                     help="per-electron tensor result size (0 = no-op)")
     ap.add_argument("--concurrency", type=int, default=1)
+    ap.add_argument("--mix", action="store_true",
+                    help="mixed workload: no-ops, bf16 matmuls, tensor returns, "
+                         "exceptions (GPU electrons when cuda is visible)")
     args = ap.parse_args()
 
-    if args.tensor_bytes:
+    if args.mix:
+
+        def electron(i):
+            kind = i % 5
+            if kind == 0:
+                return i
+            if kind == 1:
+                raise ValueError(f"intentional {i}")
+            import torch
+
+            dev = "cuda" if torch.cuda.is_available() else "cpu"
+            if kind == 2:  # matmul
+                a = torch.randn(512, 512, device=dev, dtype=torch.bfloat16)
+                c = a @ a
+                if dev == "cuda":
+                    torch.cuda.synchronize()
+                return float(c.float().mean())
+            if kind == 3:  # tensor return through staging
+                return torch.ones(4 << 20, device=dev, dtype=torch.bfloat16)
+            return {"nested": [torch.arange(1000), (i, "tag")]}
+
+        fargs = None  # per-task arg
+    elif args.tensor_bytes:
 
         def electron(nbytes):
             import torch
@@ -85,10 +110,20 @@ def main() -> None:
             t0 = time.perf_counter()
             sem = asyncio.Semaphore(args.concurrency)
 
+            errors = [0]
+
             async def one(i):
                 async with sem:
-                    await ex.execute(electron, list(fargs), {},
-                                     dispatch_id="soak", node_id=i)
+                    try:
+                        await ex.execute(
+                            electron,
+                            [i] if fargs is None else list(fargs),
+                            {},
+                            dispatch_id="soak",
+                            node_id=i,
+                        )
+                    except ValueError:
+                        errors[0] += 1  # the mix's intentional failures
 
             done = 0
             worker_pid = None
@@ -107,7 +142,8 @@ def main() -> None:
                 )
             elapsed = time.perf_counter() - t0
             print(f"TOTAL {args.electrons} electrons in {elapsed:.1f}s = "
-                  f"{args.electrons/elapsed:.1f} e/s")
+                  f"{args.electrons/elapsed:.1f} e/s; "
+                  f"intentional-failure roundtrips: {errors[0]}")
             await SSHExecutor.close_pool()
 
     asyncio.run(run())
