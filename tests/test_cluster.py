@@ -73,3 +73,10 @@ def test_cluster_stats_and_capacity(tmp_path):
     stats = cluster.stats()
     assert len(stats) == 2
     assert sum(s["counters"]["tasks"] for s in stats.values()) == 1
+
+
+def test_cluster_prewarm(tmp_path):
+    cluster, _ = _cluster(tmp_path, persistent_workers=True, cpu_workers=2)
+    n = asyncio.run(cluster.prewarm())
+    assert n == 4  # 2 hosts x 2 cpu workers
+    assert asyncio.run(cluster.execute(lambda: "warm", [], {})) == "warm"
