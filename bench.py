@@ -185,8 +185,11 @@ def main() -> None:
             lat = [await one_step() for _ in range(args.steps)]
             barrier_sync()
             elapsed = time.perf_counter() - t_start
+            from covalent_ssh_plugin_amd.utils.timing import summarize
+
+            phase_means = summarize(ex.task_records).get("phase_mean_ms", {})
             await SSHExecutor.close_pool()
-            return {"elapsed": elapsed, "latencies": lat}
+            return {"elapsed": elapsed, "latencies": lat, "phases": phase_means}
 
     out = asyncio.run(run_bench())
     elapsed = out["elapsed"]
@@ -227,6 +230,7 @@ def main() -> None:
                 + ("persistent-worker dispatch" if args.config != "noop-stub" else "spawn-per-task stub dispatch")
             ),
             "p50_ms": round(p50_ms, 3),
+            "phase_mean_ms": {k: round(v, 3) for k, v in out.get("phases", {}).items()},
             "config": {
                 "model": f"{args.config}-electron-dispatch",
                 "global_batch": world_size,

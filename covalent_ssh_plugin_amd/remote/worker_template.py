@@ -91,6 +91,7 @@ import cloudpickle as pickle  # noqa: E402  (after fd surgery on purpose)
 
 _gpu_lib = None
 _gpu_info = None
+_served = 0  # electrons completed by this worker (observability)
 
 
 def _ensure_torch_runtime_first():
@@ -287,11 +288,13 @@ def _rebuild_arg_tensors(obj, buffers, meta_list):
 
 
 def _serve_one(request):
+    global _served
+    _served += 1
     t0 = time.monotonic()
     meta = {"phases_ms": {}, "gpu": _gpu_info, "staging": None,
             "gpu_slot": GPU_SLOT, "buffers": [],
             "hip_visible_devices": os.environ.get("HIP_VISIBLE_DEVICES"),
-            "pid": os.getpid(), "worker": True}
+            "pid": os.getpid(), "worker": True, "served": _served}
     result = None
     exception = None
     buffers = []
