@@ -16,7 +16,7 @@ python + HIP-runtime start per electron (the reference's architecture,
     ``(result, exception)`` 2-tuple (same payload contract as the
     result file) plus a meta JSON dict.
 
-Protocol (all frames: 4-byte big-endian length + payload):
+Protocol (all frames: 8-byte big-endian length + payload):
   request  = pickle dict {"op_id", "workdir", "function_blob",
              "arg_buffers": [{dtype, shape}, ...]} followed by one RAW
              frame per entry (large CPU-tensor arguments, mirrored from
@@ -236,12 +236,12 @@ def _read_frame(fd, idle_timeout=0.0):
         if not ready:
             return None  # idle too long: orderly exit
     header = b""
-    while len(header) < 4:
-        chunk = os.read(fd, 4 - len(header))
+    while len(header) < 8:
+        chunk = os.read(fd, 8 - len(header))
         if not chunk:
             return None
         header += chunk
-    (length,) = struct.unpack(">I", header)
+    (length,) = struct.unpack(">Q", header)
     if length == 0:
         return b""
     parts = []
@@ -256,7 +256,7 @@ def _read_frame(fd, idle_timeout=0.0):
 
 
 def _write_frame(fd, payload):
-    os.write(fd, struct.pack(">I", len(payload)))
+    os.write(fd, struct.pack(">Q", len(payload)))
     view = memoryview(payload)
     while view:
         written = os.write(fd, view[: 1 << 20])

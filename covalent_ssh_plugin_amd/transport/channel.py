@@ -2,7 +2,8 @@
 
 A :class:`Channel` wraps one persistent remote process (a worker,
 remote/worker_template.py) whose stdin/stdout carry 4-byte
-length-prefixed frames.  Channels are how persistent workers are fed:
+length-prefixed frames (8-byte big-endian lengths: single tensor
+buffers can exceed 4 GiB on a 288 GB HBM3E node).  Channels are how persistent workers are fed:
 over SSH the process rides the pooled ControlMaster (no per-task
 handshake), locally it is a plain subprocess — either way the remote
 python + HIP runtime stay warm across electrons.
@@ -41,14 +42,14 @@ class Channel:
         if not self.alive:
             raise ChannelClosed(f"{self._label}: process exited")
         view = memoryview(payload).cast("B")
-        self._proc.stdin.write(struct.pack(">I", view.nbytes))
+        self._proc.stdin.write(struct.pack(">Q", view.nbytes))
         self._proc.stdin.write(view)
         await self._proc.stdin.drain()
 
     async def recv_frame(self, timeout: Optional[float] = None) -> bytes:
         async def _read() -> bytes:
-            header = await self._proc.stdout.readexactly(4)
-            (length,) = struct.unpack(">I", header)
+            header = await self._proc.stdout.readexactly(8)
+            (length,) = struct.unpack(">Q", header)
             if length == 0:
                 return b""
             if length <= (4 << 20):
@@ -60,7 +61,12 @@ class Channel:
             view = memoryview(buf)
             pos = 0
             while pos < length:
-                chunk = await self._proc.stdout.read(min(1 << 20, length - pos))
+                # read(remaining): takes the reader's ENTIRE internal
+                # buffer in one pop.  Small fixed-size reads here are
+                # quadratic — StreamReader deletes consumed bytes from
+                # the front of its (up to 2*limit = 128 MiB) buffer, an
+                # O(buffer) memmove per call.
+                chunk = await self._proc.stdout.read(length - pos)
                 if not chunk:
                     raise asyncio.IncompleteReadError(bytes(view[:pos]), length)
                 view[pos : pos + len(chunk)] = chunk
@@ -191,7 +197,7 @@ class Channel:
         if self.alive:
             try:
                 # zero-length frame = orderly shutdown
-                self._proc.stdin.write(struct.pack(">I", 0))
+                self._proc.stdin.write(struct.pack(">Q", 0))
                 await self._proc.stdin.drain()
                 self._proc.stdin.close()
             except (ConnectionResetError, BrokenPipeError, RuntimeError):

@@ -246,3 +246,32 @@ def test_pinned_h2d_roundtrip(gpu_lib):
         assert torch.equal(dev.cpu(), src)
     finally:
         gpu_lib.csp_host_free(ctypes.c_void_p(host))
+
+
+@pytest.mark.timeout(600)
+def test_worker_tensor_over_4gib(tmp_path):
+    """Single result tensors beyond a 4-byte frame-length limit (the
+    protocol uses 8-byte lengths; 288 GB HBM makes multi-GiB results
+    routine).  Runs in the GPU tier because CI microVMs page large
+    allocations erratically; any real machine finishes in seconds."""
+    import asyncio
+
+    ex = _executor(tmp_path, persistent_workers=True,
+                   pinned_staging_threshold_bytes=1 << 20, warmup_gpu=False)
+
+    n = (4 * 1024 + 512) * 1024 * 1024  # 4.5 GiB of uint8
+
+    def fn(n):
+        import torch
+
+        t = torch.zeros(n, dtype=torch.uint8, device="cuda")
+        t[0] = 7
+        t[-1] = 9
+        torch.cuda.synchronize()
+        return t
+
+    out = asyncio.run(ex.execute(fn, [n], {}, dispatch_id="big", node_id=0))
+    assert out.numel() == n
+    assert int(out[0]) == 7 and int(out[-1]) == 9
+    assert ex.last_task_record.remote_meta["staging"]["mode"] == "pinned"
+    asyncio.run(ex.close_pool())

@@ -119,11 +119,11 @@ def test_channel_frame_roundtrip(tmp_path_factory, payloads):
         "proto = os.dup(1); os.dup2(2, 1)\n"
         "def rf():\n"
         "    h = b''\n"
-        "    while len(h) < 4:\n"
-        "        c = os.read(0, 4 - len(h))\n"
+        "    while len(h) < 8:\n"
+        "        c = os.read(0, 8 - len(h))\n"
         "        if not c: return None\n"
         "        h += c\n"
-        "    (n,) = struct.unpack('>I', h)\n"
+        "    (n,) = struct.unpack('>Q', h)\n"
         "    data = b''\n"
         "    while len(data) < n:\n"
         "        c = os.read(0, n - len(data))\n"
@@ -133,7 +133,7 @@ def test_channel_frame_roundtrip(tmp_path_factory, payloads):
         "while True:\n"
         "    d = rf()\n"
         "    if d is None or d == b'': break\n"
-        "    os.write(proto, struct.pack('>I', len(d))); os.write(proto, d)\n"
+        "    os.write(proto, struct.pack('>Q', len(d))); os.write(proto, d)\n"
     )
 
     async def main():

@@ -500,12 +500,12 @@ def test_worker_standalone_protocol_and_slot_resolution(tmp_path):
     )
 
     def send(payload: bytes):
-        proc.stdin.write(struct.pack(">I", len(payload)) + payload)
+        proc.stdin.write(struct.pack(">Q", len(payload)) + payload)
         proc.stdin.flush()
 
     def recv() -> bytes:
-        header = proc.stdout.read(4)
-        (n,) = struct.unpack(">I", header)
+        header = proc.stdout.read(8)
+        (n,) = struct.unpack(">Q", header)
         return proc.stdout.read(n)
 
     try:
