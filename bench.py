@@ -91,8 +91,11 @@ def build_electron(config: str):
 def main() -> None:
     parser = argparse.ArgumentParser()
     parser.add_argument("--gpus", type=int, default=1)
-    parser.add_argument("--steps", type=int, default=256)
-    parser.add_argument("--warmup", type=int, default=32)
+    # defaults sized so the timed region is long enough to amortize
+    # singleton OS scheduling hiccups (~30 ms) that otherwise dominate
+    # a sub-second measurement; still finishes in seconds
+    parser.add_argument("--steps", type=int, default=2048)
+    parser.add_argument("--warmup", type=int, default=64)
     parser.add_argument("--config", default="noop")
     parser.add_argument("--dump-latencies", default="",
                         help="write per-step latencies (seconds, one per line)")
