@@ -94,14 +94,20 @@ def main() -> None:
     # defaults sized so the timed region is long enough to amortize
     # singleton OS scheduling hiccups (~30 ms) that otherwise dominate
     # a sub-second measurement; still finishes in seconds
-    parser.add_argument("--steps", type=int, default=2048)
-    parser.add_argument("--warmup", type=int, default=64)
+    parser.add_argument("--steps", type=int, default=None)
+    parser.add_argument("--warmup", type=int, default=None)
     parser.add_argument("--config", default="noop")
     parser.add_argument("--dump-latencies", default="",
                         help="write per-step latencies (seconds, one per line)")
     parser.add_argument("--fan", type=int, default=64,
                         help="concurrent electrons per step for --config fan")
     args = parser.parse_args()
+    if args.steps is None:
+        # per-config defaults: long enough to amortize ~30 ms OS hiccups
+        # for the fast configs, bounded wall time for the heavy ones
+        args.steps = {"staging": 8, "rccl": 3, "fan": 64}.get(args.config, 2048)
+    if args.warmup is None:
+        args.warmup = {"staging": 2, "rccl": 1, "fan": 4}.get(args.config, 64)
 
     import torch
 
