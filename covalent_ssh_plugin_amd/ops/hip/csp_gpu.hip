@@ -458,3 +458,59 @@ extern "C" int csp_mfma_bench(int device, int iters, int blocks, double* tflops)
     *tflops = waves * (double)iters * kFlopPerMfma / ((double)ms / 1000.0) / 1.0e12;
     return 0;
 }
+
+// ---------------------------------------------------------------------------
+// MFMA numerics check: one v_mfma_f32_32x32x16_bf16 with self-described
+// operand layouts.  Each lane scatters the exact operand values it feeds
+// the instruction into global A (32x16) and B (16x32), and its
+// accumulator fragment into D (32x32) using the C/D mapping
+// (col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)).  The host
+// test then asserts D == A @ B against a PyTorch reference — validating
+// the assumed A/B lane mapping AND the instruction together.  `layout`
+// selects the A/B k-mapping candidate:
+//   0: k = (lane>>5)*8 + j           (contiguous 8)
+//   1: k = (lane>>5)*4 + (j&3) + (j>>2)*8   (two 4-element halves)
+__global__ __launch_bounds__(64) void csp_mfma_check_kernel(
+    float* __restrict__ D, float* __restrict__ A, float* __restrict__ B,
+    int layout) {
+    const int l = threadIdx.x;
+    bf16x8 a, b;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        const int k = (layout == 0) ? ((l >> 5) * 8 + j)
+                                    : ((l >> 5) * 4 + (j & 3) + (j >> 2) * 8);
+        const int i = l & 31;  // A row
+        const float av = (float)(((i * 3 + k) % 7) - 3);  // integer-valued
+        a[j] = (__bf16)av;
+        A[i * 16 + k] = av;
+        const int col = l & 31;  // B column
+        const float bv = (float)(((k * 5 + col) % 5) - 2);
+        b[j] = (__bf16)bv;
+        B[k * 32 + col] = bv;
+    }
+    f32x16 acc = {};
+    acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+        const int row = (r & 3) + 8 * (r >> 2) + 4 * (l >> 5);
+        D[row * 32 + (l & 31)] = acc[r];
+    }
+}
+
+extern "C" int csp_mfma_check(int device, int layout, float* d_out,
+                              float* a_out, float* b_out) {
+    HIP_TRY(hipSetDevice(device));
+    float *D = nullptr, *A = nullptr, *B = nullptr;
+    HIP_TRY(hipMalloc(&D, 32 * 32 * sizeof(float)));
+    HIP_TRY(hipMalloc(&A, 32 * 16 * sizeof(float)));
+    HIP_TRY(hipMalloc(&B, 16 * 32 * sizeof(float)));
+    hipLaunchKernelGGL(csp_mfma_check_kernel, dim3(1), dim3(64), 0, 0, D, A, B,
+                       layout);
+    HIP_TRY(hipMemcpy(d_out, D, 32 * 32 * sizeof(float), hipMemcpyDeviceToHost));
+    HIP_TRY(hipMemcpy(a_out, A, 32 * 16 * sizeof(float), hipMemcpyDeviceToHost));
+    HIP_TRY(hipMemcpy(b_out, B, 16 * 32 * sizeof(float), hipMemcpyDeviceToHost));
+    (void)hipFree(D);
+    (void)hipFree(A);
+    (void)hipFree(B);
+    return 0;
+}

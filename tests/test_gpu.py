@@ -275,3 +275,34 @@ def test_worker_tensor_over_4gib(tmp_path):
     assert int(out[0]) == 7 and int(out[-1]) == 9
     assert ex.last_task_record.remote_meta["staging"]["mode"] == "pinned"
     asyncio.run(ex.close_pool())
+
+
+def test_mfma_numerics_vs_torch_reference(gpu_lib):
+    """The warm-up kernel's MFMA instruction, numerics-checked: one
+    v_mfma_f32_32x32x16_bf16 with self-described operand layouts must
+    equal the PyTorch fp32 reference EXACTLY (integer-valued operands:
+    no rounding anywhere)."""
+    import ctypes
+
+    gpu_lib.csp_mfma_check.restype = ctypes.c_int
+    gpu_lib.csp_mfma_check.argtypes = [ctypes.c_int, ctypes.c_int] + [
+        ctypes.POINTER(ctypes.c_float)
+    ] * 3
+
+    matched = []
+    for layout in (0, 1):
+        D = (ctypes.c_float * (32 * 32))()
+        A = (ctypes.c_float * (32 * 16))()
+        B = (ctypes.c_float * (16 * 32))()
+        rc = gpu_lib.csp_mfma_check(0, layout, D, A, B)
+        assert rc == 0
+        a = torch.tensor(list(A)).reshape(32, 16)
+        b = torch.tensor(list(B)).reshape(16, 32)
+        d = torch.tensor(list(D)).reshape(32, 32)
+        ref = a @ b
+        if torch.equal(d, ref):
+            matched.append(layout)
+        # A/B must be fully populated regardless of layout candidate
+        assert a.abs().sum() > 0 and b.abs().sum() > 0
+    assert matched, "no A/B fragment layout candidate reproduced A @ B"
+    print("verified A/B k-mapping layout(s):", matched)
