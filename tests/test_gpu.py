@@ -304,5 +304,10 @@ def test_mfma_numerics_vs_torch_reference(gpu_lib):
             matched.append(layout)
         # A/B must be fully populated regardless of layout candidate
         assert a.abs().sum() > 0 and b.abs().sum() > 0
-    assert matched, "no A/B fragment layout candidate reproduced A @ B"
+    # NOTE: a k-index permutation applied to BOTH A and B leaves
+    # sum_k A[i,k]*B[k,j] invariant, so both candidates pass (verified
+    # on-box: [0, 1]).  What the test pins down is the numerics contract:
+    # the instruction computes exactly the matrix product of the operands
+    # as fed, bit-equal to the fp32 reference.
+    assert matched, "MFMA output did not reproduce A @ B"
     print("verified A/B k-mapping layout(s):", matched)
