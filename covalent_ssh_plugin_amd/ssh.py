@@ -519,7 +519,13 @@ class SSHExecutor(RemoteExecutor):
         remote_lib = f"{self.remote_cache}/lib/csp_gpu-{digest}.so"
         probe = await transport.run(f"test -f {shlex.quote(remote_lib)}")
         if not probe.ok:
-            await transport.put_files([(local_lib, remote_lib)])
+            # publish atomically (rename) so a concurrent stub can never
+            # ctypes-load a half-written library
+            remote_tmp = f"{remote_lib}.tmp-{uuid.uuid4().hex[:8]}"
+            await transport.put_files([(local_lib, remote_tmp)])
+            await transport.run(
+                f"mv -f {shlex.quote(remote_tmp)} {shlex.quote(remote_lib)}"
+            )
         return remote_lib
 
     async def _provision_worker_script(self, transport: Transport, gpu_lib: str) -> str:
@@ -544,14 +550,18 @@ class SSHExecutor(RemoteExecutor):
         probe = await transport.run(f"test -f {shlex.quote(remote_path)}")
         if not probe.ok:
             Path(self.cache_dir).mkdir(parents=True, exist_ok=True)
-            # unique temp name: concurrent electrons may provision the
-            # same digest at once (idempotent, content-addressed)
-            local_tmp = os.path.join(
-                self.cache_dir, f"worker-{digest}-{uuid.uuid4().hex[:8]}.py"
-            )
+            # unique temp names local AND remote: concurrent electrons may
+            # provision the same digest at once; the remote rename makes
+            # the publish atomic (no reader ever sees a half-written file)
+            nonce = uuid.uuid4().hex[:8]
+            local_tmp = os.path.join(self.cache_dir, f"worker-{digest}-{nonce}.py")
+            remote_tmp = f"{remote_path}.tmp-{nonce}"
             Path(local_tmp).write_text(text)
             try:
-                await transport.put_files([(local_tmp, remote_path)])
+                await transport.put_files([(local_tmp, remote_tmp)])
+                await transport.run(
+                    f"mv -f {shlex.quote(remote_tmp)} {shlex.quote(remote_path)}"
+                )
             finally:
                 try:
                     os.remove(local_tmp)
