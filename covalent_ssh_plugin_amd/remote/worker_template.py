@@ -21,6 +21,13 @@ Protocol (all frames: 8-byte big-endian length + payload):
              "arg_buffers": [{dtype, shape}, ...]} followed by one RAW
              frame per entry (large CPU-tensor arguments, mirrored from
              the dispatcher without pickle copies)
+  ack      = pickle tuple ("A1", op_id) — written as soon as the request
+             (and its arg frames) has been fully received, BEFORE any
+             user code runs.  The dispatcher uses it to distinguish
+             "worker died before starting the task" (safe to retry on a
+             fresh worker) from "worker died mid-execution" (retrying
+             would re-run a possibly non-idempotent task — surfaced as
+             an error instead, unless retry_on_worker_death opts in)
   response = pickle tuple ("R1", result_blob: bytes, meta: dict, nbuf: int)
              followed by nbuf RAW tensor-buffer frames.  Large tensors in
              the result are replaced by ("__csp_tensor_buffer_v1__", i)
@@ -388,6 +395,9 @@ def main():
                     return
                 frames.append(bytearray(f))
             request["_arg_buffer_frames"] = frames
+        # ack AFTER the full request is in hand, BEFORE user code runs:
+        # the dispatcher's retry-on-death policy keys off this frame
+        _write_frame(_proto_fd, pickle.dumps(("A1", request.get("op_id"))))
         result_blob, meta, buffers = _serve_one(request)
         _write_frame(_proto_fd, pickle.dumps(("R1", result_blob, meta, len(buffers))))
         for view, _keep in buffers:

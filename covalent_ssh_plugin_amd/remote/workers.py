@@ -195,8 +195,15 @@ async def run_task(
     timeout: Optional[float] = None,
     arg_buffer_meta=None,
     arg_buffers=None,
+    ack_state: Optional[dict] = None,
 ):
-    """One electron through a worker.  Returns (result, exception, meta)."""
+    """One electron through a worker.  Returns (result, exception, meta).
+
+    ``ack_state`` (a mutable dict) gets ``{"started": True}`` the moment
+    the worker's A1 ack arrives — i.e. once user code may have begun.
+    On ChannelClosed the caller reads it to decide whether a retry could
+    re-execute a partially-run task.
+    """
     request = cloudpickle.dumps(
         {
             "op_id": op_id,
@@ -208,8 +215,13 @@ async def run_task(
     frames = [request] + [view for view, _keep in (arg_buffers or [])]
     parsed = {}
 
-    def reply_frames(main: bytes) -> int:
-        tag, result_blob, meta, nbuf = cloudpickle.loads(main)
+    def reply_frames(main: bytes):
+        decoded = cloudpickle.loads(main)
+        if decoded[0] == "A1":
+            if ack_state is not None:
+                ack_state["started"] = True
+            return None  # control frame: the R1 reply follows
+        tag, result_blob, meta, nbuf = decoded
         assert tag == "R1", f"unexpected worker response tag {tag!r}"
         parsed["value"] = (result_blob, meta, nbuf)
         return nbuf
@@ -219,6 +231,7 @@ async def run_task(
     _main, buffers = await handle.channel.exchange(
         frames, reply_frames, timeout=timeout
     )
+    assert "value" in parsed, "worker reply ended at the ack frame"
     result_blob, meta, nbuf = parsed["value"]
     result, exception = cloudpickle.loads(result_blob)
     if nbuf:

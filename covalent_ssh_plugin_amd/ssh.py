@@ -77,7 +77,12 @@ _EXECUTOR_PLUGIN_DEFAULTS = {
     "ssh_port": 22,
     "gpu_slots": 8,  # GPUs on the remote node (8×MI355X)
     "slots_per_gpu": 1,  # 288 GB HBM3E/GPU: 1 electron per GPU by default
-    "hip_visible_devices_policy": "roundrobin",  # "roundrobin" | "none"
+    # "roundrobin": each task ACQUIRES a slot (bounded concurrency, FIFO
+    #   fair, spreads across GPUs).  "fixed": every task pins to
+    #   `fixed_gpu` WITHOUT slot accounting — unbounded concurrency on
+    #   that one GPU, intended for per-rank bench/embarrassing-parallel
+    #   drivers that do their own admission control.  "none": no pinning.
+    "hip_visible_devices_policy": "roundrobin",
     "warmup_gpu": True,  # run the CDNA4 warm-up/probe kernel pre-task
     "pinned_staging_threshold_bytes": DEFAULT_STAGING_THRESHOLD,
     "batch_roundtrips": True,  # fused single-round-trip dispatch
@@ -85,6 +90,11 @@ _EXECUTOR_PLUGIN_DEFAULTS = {
     "cpu_workers": 4,  # worker-set size when no GPU policy is active
     "task_timeout": 0,  # seconds; 0 = unlimited
     "worker_idle_timeout": 0,  # seconds; 0 = workers never exit on idle
+    # Opt-in: re-run a task whose worker died AFTER execution had started
+    # (the worker acks each request before running user code; a task that
+    # never acked is always safe to retry and always is).  Off by default
+    # because a non-idempotent task could execute twice.
+    "retry_on_worker_death": False,
 }
 
 update_config_defaults("executors.ssh", _EXECUTOR_PLUGIN_DEFAULTS)
@@ -152,6 +162,7 @@ class SSHExecutor(RemoteExecutor):
         cpu_workers: Optional[int] = None,
         task_timeout: Optional[float] = None,
         worker_idle_timeout: Optional[float] = None,
+        retry_on_worker_death: Optional[bool] = None,
         fixed_gpu: int = 0,
         ssh_extra_options: Optional[List[str]] = None,
         local_home: str = "",
@@ -200,6 +211,9 @@ class SSHExecutor(RemoteExecutor):
         self.task_timeout = float(_conf("task_timeout", task_timeout, default=0) or 0)
         self.worker_idle_timeout = float(
             _conf("worker_idle_timeout", worker_idle_timeout, default=0) or 0
+        )
+        self.retry_on_worker_death = bool(
+            _conf("retry_on_worker_death", retry_on_worker_death, default=False)
         )
         self.fixed_gpu = int(fixed_gpu)
         #: raw `ssh -o`/flag passthrough, e.g. ["-o", "StrictHostKeyChecking=yes",
@@ -376,8 +390,13 @@ class SSHExecutor(RemoteExecutor):
         Here, a task dispatched through a persistent worker CAN be
         cancelled: the worker process serving it is killed (the in-flight
         request fails with ChannelClosed and a fresh worker replaces it
-        for subsequent electrons).  Stub-dispatched tasks remain
-        uncancellable, matching the reference.
+        for subsequent electrons).  If OTHER electrons were pipelined on
+        the same worker, any that had already started execution fail
+        loudly with an explanatory SSHTaskError instead of being silently
+        re-executed (worker A1 ack protocol); electrons the worker had
+        not yet started are transparently re-dispatched — always safe.
+        Stub-dispatched tasks remain uncancellable, matching the
+        reference.
         """
         task_metadata = task_metadata or {}
         operation_id = (
@@ -444,27 +463,33 @@ class SSHExecutor(RemoteExecutor):
     # per-task path; reference does these per task at ssh.py:508-532)
     # ------------------------------------------------------------------
 
-    async def _ensure_environment(self, transport: Transport) -> str:
+    async def _ensure_environment(self, transport: Transport) -> Tuple[str, bool]:
         """Verify python/conda once per endpoint and provision the GPU
-        library.  Returns the remote GPU library path ('' if absent)."""
+        library.  Returns ``(remote_gpu_lib_path_or_empty, has_gpu)`` —
+        the two are independent: slot scheduling keys off ``has_gpu``
+        (the endpoint exposes /dev/kfd), library provisioning only adds
+        the warm-up/pinned-staging fast path (pinning must
+        not silently disappear when the dispatcher-side .so is absent)."""
         key = self._pool_key()
         cached = transport_pool.cached_check(key, "env")
         if cached is not None:
-            ok, detail, gpu_lib = cached
+            ok, detail, gpu_lib, has_gpu = cached
             if not ok:
                 raise SSHTaskError(detail)
-            return gpu_lib
+            return gpu_lib, has_gpu
 
         async with transport_pool.check_lock(key):
             cached = transport_pool.cached_check(key, "env")
             if cached is not None:
-                ok, detail, gpu_lib = cached
+                ok, detail, gpu_lib, has_gpu = cached
                 if not ok:
                     raise SSHTaskError(detail)
-                return gpu_lib
+                return gpu_lib, has_gpu
             return await self._ensure_environment_locked(transport, key)
 
-    async def _ensure_environment_locked(self, transport: Transport, key) -> str:
+    async def _ensure_environment_locked(
+        self, transport: Transport, key
+    ) -> Tuple[str, bool]:
         # conda env existence (reference ssh.py:508-519)
         if self.conda_env:
             proc = await transport.run(
@@ -472,7 +497,7 @@ class SSHExecutor(RemoteExecutor):
             )
             if not proc.ok:
                 detail = f"conda environment {self.conda_env!r} not found on {transport.endpoint}"
-                transport_pool.store_check(key, "env", (False, detail, ""))
+                transport_pool.store_check(key, "env", (False, detail, "", False), ok=False)
                 raise SSHTaskError(detail)
 
         # python3 sanity (reference ssh.py:521-524) + remote cache dir +
@@ -487,15 +512,23 @@ class SSHExecutor(RemoteExecutor):
                 f"no python3 at {self.python_path!r} on {transport.endpoint}: "
                 f"{proc.text_err().strip()}"
             )
-            transport_pool.store_check(key, "env", (False, detail, ""))
+            transport_pool.store_check(key, "env", (False, detail, "", False), ok=False)
             raise SSHTaskError(detail)
 
-        # Only provision the CDNA4 library (and later inject GPU slots)
-        # when the endpoint actually has an AMD GPU stack.
+        # Only provision the CDNA4 library when the endpoint actually has
+        # an AMD GPU stack; slot scheduling follows has_gpu alone.
         has_gpu = "CSP_HAS_GPU" in proc.text_out()
         gpu_lib = await self._provision_gpu_lib(transport) if has_gpu else ""
-        transport_pool.store_check(key, "env", (True, "", gpu_lib))
-        return gpu_lib
+        if has_gpu and not gpu_lib:
+            app_log.warning(
+                "endpoint %s has a GPU stack but the CDNA4 warm-up/staging "
+                "library (libcsp_gpu.so) is not built on the dispatcher — "
+                "GPU slot pinning stays ACTIVE, but tasks run without the "
+                "warm-up kernel and pinned staging fast path",
+                transport.endpoint,
+            )
+        transport_pool.store_check(key, "env", (True, "", gpu_lib, has_gpu))
+        return gpu_lib, has_gpu
 
     def _wrap_conda(self, cmd: str) -> str:
         if self.conda_env:
@@ -611,6 +644,7 @@ class SSHExecutor(RemoteExecutor):
                         raise
                     await asyncio.sleep(0.5)
                     continue
+                ack_state = {"started": False}
                 try:
                     return await worker_pool.run_task(
                         handle,
@@ -620,6 +654,7 @@ class SSHExecutor(RemoteExecutor):
                         timeout=self.task_timeout or None,
                         arg_buffer_meta=arg_meta,
                         arg_buffers=arg_bufs,
+                        ack_state=ack_state,
                     )
                 except asyncio.TimeoutError:
                     # the worker is wedged on this task: kill it so the
@@ -634,11 +669,29 @@ class SSHExecutor(RemoteExecutor):
                     if operation_id in self._cancelled:
                         self._cancelled.discard(operation_id)
                         raise SSHTaskError(f"task {operation_id} was cancelled")
+                    # Re-dispatch policy: the worker acks each
+                    # request before executing user code.  No ack => the
+                    # task never started and a retry cannot re-execute
+                    # anything.  Ack received => user code may have
+                    # partially run; re-running a non-idempotent task is
+                    # only done when the user opted in.
+                    if ack_state["started"] and not self.retry_on_worker_death:
+                        raise SSHTaskError(
+                            f"worker serving task {operation_id} died after "
+                            "execution had started (possibly killed by a "
+                            "cancel of a co-resident electron); not "
+                            "re-executing automatically — the task may have "
+                            "partially run.  Set retry_on_worker_death=True "
+                            "to opt in to re-execution of idempotent tasks."
+                        )
                     if attempt == 1:
                         raise
                     self.counters["worker_respawns"] += 1
                     app_log.warning(
-                        "worker %s died mid-task; respawning once", key
+                        "worker %s died %s task %s; respawning once",
+                        key,
+                        "during" if ack_state["started"] else "before",
+                        operation_id,
                     )
         finally:
             self._inflight.pop(operation_id, None)
@@ -761,12 +814,15 @@ class SSHExecutor(RemoteExecutor):
 
         with timer.phase("env_checks"):
             try:
-                gpu_lib = await self._ensure_environment(transport)
+                gpu_lib, has_gpu = await self._ensure_environment(transport)
             except SSHTaskError as e:
                 return await self._on_ssh_fail(function, args, kwargs, str(e))
 
         slot = None
-        policy = self.hip_visible_devices_policy if gpu_lib else "none"
+        # Slot policy keys off GPU PRESENCE, not library provisioning:
+        # concurrency limiting + HIP_VISIBLE_DEVICES pinning must hold
+        # even when the warm-up .so is unavailable.
+        policy = self.hip_visible_devices_policy if has_gpu else "none"
         try:
             if policy == "roundrobin":
                 with timer.phase("slot_wait"):
@@ -964,18 +1020,18 @@ class SSHExecutor(RemoteExecutor):
         transport = await self._client_connect()
         if transport is None:
             raise RuntimeError(f"could not connect to {self.hostname}")
-        gpu_lib = await self._ensure_environment(transport)
+        gpu_lib, has_gpu = await self._ensure_environment(transport)
         if not self.persistent_workers:
             return 0
         script_remote = await self._provision_worker_script(transport, gpu_lib)
         cmd = self._wrap_conda(f"{self.python_path} {shlex.quote(script_remote)}")
         started = 0
-        if self.hip_visible_devices_policy == "roundrobin" and gpu_lib:
+        if self.hip_visible_devices_policy == "roundrobin" and has_gpu:
             tags_envs = [
                 (gpu, {"CSP_GPU_SLOT": str(gpu)})
                 for gpu in range(slots if slots is not None else self.gpu_slots)
             ]
-        elif self.hip_visible_devices_policy == "fixed" and gpu_lib:
+        elif self.hip_visible_devices_policy == "fixed" and has_gpu:
             tags_envs = [(self.fixed_gpu, {"CSP_GPU_SLOT": str(self.fixed_gpu)})]
         else:
             tags_envs = [

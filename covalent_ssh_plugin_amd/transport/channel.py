@@ -125,6 +125,11 @@ class Channel:
                 fut, reply_frames = await self._waiters.get()
                 main = await self.recv_frame()
                 extra_count = reply_frames(main)
+                while extra_count is None:
+                    # control frame (e.g. the worker's A1 execution ack)
+                    # consumed by the callback: the real reply follows
+                    main = await self.recv_frame()
+                    extra_count = reply_frames(main)
                 extras = [await self.recv_frame() for _ in range(extra_count)]
                 if not fut.done():
                     fut.set_result((main, extras))
@@ -152,7 +157,8 @@ class Channel:
 
         ``request_frames``: iterable of payloads sent back-to-back under
         the write lock.  ``reply_frames``: callable(main_frame) -> number
-        of extra raw frames to read for this reply.  Returns
+        of extra raw frames to read for this reply, or ``None`` to mean
+        "that was a control frame; keep reading".  Returns
         ``(main_frame, [extra_frames])``.  Multiple callers may have
         requests in flight concurrently; replies resolve in order.
         """

@@ -16,17 +16,26 @@ accounting (SURVEY.md §7 "Hard parts").
 from __future__ import annotations
 
 import asyncio
+import time
 from typing import Callable, Dict, Optional, Tuple
 
 from .base import Transport
 
 PoolKey = Tuple[str, ...]
 
+#: how long a FAILED check result stays cached (seconds).  Successful
+#: checks are cached for the life of the process (the hoisting win);
+#: failures must heal: a transient remote hiccup (conda glitch, disk
+#: full during mkdir) may be gone on the next attempt (ADVICE r1:
+#: "failed environment checks are cached forever").
+NEGATIVE_CHECK_TTL = 60.0
+
 _pool: Dict[PoolKey, Transport] = {}
 _pool_lock: Optional[asyncio.Lock] = None
 _pool_lock_loop = None
-# One-time environment check results, keyed by (pool_key, check_name).
-_env_checks: Dict[Tuple[PoolKey, str], object] = {}
+# One-time environment check results, keyed by (pool_key, check_name):
+# (value, monotonic_timestamp, ok)
+_env_checks: Dict[Tuple[PoolKey, str], Tuple[object, float, bool]] = {}
 
 
 def _lock() -> asyncio.Lock:
@@ -60,7 +69,10 @@ async def get_transport(key: PoolKey, factory: Callable[[], Transport]) -> Trans
     """Return the pooled, connected transport for ``key``, creating it
     with ``factory`` (and connecting it) on first use.
 
-    If the pooled transport has dropped its connection, reconnect it.
+    If the pooled transport has dropped its connection, reconnect it —
+    and on a successful RE-connect, drop any cached FAILED environment
+    checks for the endpoint: the failure may have died with the old
+    connection (e.g. the remote host was rebooted/fixed).
     """
     async with _lock():
         transport = _pool.get(key)
@@ -68,16 +80,38 @@ async def get_transport(key: PoolKey, factory: Callable[[], Transport]) -> Trans
             transport = factory()
             _pool[key] = transport
     if not transport.is_connected:
+        reconnecting = bool(getattr(transport, "_csp_ever_connected", False))
         await transport.connect()
+        if reconnecting:
+            invalidate_failed_checks(key)
+    transport._csp_ever_connected = True
     return transport
 
 
 def cached_check(key: PoolKey, name: str):
-    return _env_checks.get((key, name))
+    """Cached check value, or None if absent / an expired failure."""
+    entry = _env_checks.get((key, name))
+    if entry is None:
+        return None
+    value, ts, ok = entry
+    if not ok and (time.monotonic() - ts) > NEGATIVE_CHECK_TTL:
+        _env_checks.pop((key, name), None)
+        return None
+    return value
 
 
-def store_check(key: PoolKey, name: str, value: object) -> None:
-    _env_checks[(key, name)] = value
+def store_check(key: PoolKey, name: str, value: object, ok: bool = True) -> None:
+    """Cache a check result.  ``ok=False`` marks it as a failure, which
+    expires after NEGATIVE_CHECK_TTL and is dropped on reconnect."""
+    _env_checks[(key, name)] = (value, time.monotonic(), ok)
+
+
+def invalidate_failed_checks(key: PoolKey) -> None:
+    """Drop every cached FAILED check for ``key`` (called on transport
+    reconnect; a fresh connection deserves a fresh verdict)."""
+    stale = [k for k, (_v, _ts, ok) in _env_checks.items() if k[0] == key and not ok]
+    for k in stale:
+        _env_checks.pop(k, None)
 
 
 async def close_all() -> None:

@@ -529,6 +529,9 @@ def test_worker_standalone_protocol_and_slot_resolution(tmp_path):
                 }
             )
         )
+        # ack frame first (execution-start marker), then the response
+        ack_tag, ack_op = cloudpickle.loads(recv())
+        assert (ack_tag, ack_op) == ("A1", "x_0")
         tag, blob, meta, nbuf = cloudpickle.loads(recv())
         assert tag == "R1" and nbuf == 0
         result, exception = cloudpickle.loads(blob)
