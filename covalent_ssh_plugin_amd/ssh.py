@@ -579,29 +579,39 @@ class SSHExecutor(RemoteExecutor):
         cached = transport_pool.cached_check(key, check_name)
         if cached:
             return cached
-        remote_path = f"{self.remote_cache}/lib/worker-{digest}.py"
-        probe = await transport.run(f"test -f {shlex.quote(remote_path)}")
-        if not probe.ok:
-            Path(self.cache_dir).mkdir(parents=True, exist_ok=True)
-            # unique temp names local AND remote: concurrent electrons may
-            # provision the same digest at once; the remote rename makes
-            # the publish atomic (no reader ever sees a half-written file)
-            nonce = uuid.uuid4().hex[:8]
-            local_tmp = os.path.join(self.cache_dir, f"worker-{digest}-{nonce}.py")
-            remote_tmp = f"{remote_path}.tmp-{nonce}"
-            Path(local_tmp).write_text(text)
-            try:
-                await transport.put_files([(local_tmp, remote_tmp)])
-                await transport.run(
-                    f"mv -f {shlex.quote(remote_tmp)} {shlex.quote(remote_path)}"
+        # Serialize provisioning per endpoint: a cold-start fan of N
+        # electrons must upload once, not N times — and concurrent tar
+        # extractions creating the same remote lib/ dir can race inside
+        # tar's mkdir and fail spuriously.
+        async with transport_pool.check_lock(key):
+            cached = transport_pool.cached_check(key, check_name)
+            if cached:
+                return cached
+            remote_path = f"{self.remote_cache}/lib/worker-{digest}.py"
+            probe = await transport.run(f"test -f {shlex.quote(remote_path)}")
+            if not probe.ok:
+                Path(self.cache_dir).mkdir(parents=True, exist_ok=True)
+                # unique temp names local AND remote; the remote rename
+                # makes the publish atomic (no reader ever sees a
+                # half-written file)
+                nonce = uuid.uuid4().hex[:8]
+                local_tmp = os.path.join(
+                    self.cache_dir, f"worker-{digest}-{nonce}.py"
                 )
-            finally:
+                remote_tmp = f"{remote_path}.tmp-{nonce}"
+                Path(local_tmp).write_text(text)
                 try:
-                    os.remove(local_tmp)
-                except OSError:
-                    pass
-        transport_pool.store_check(key, check_name, remote_path)
-        return remote_path
+                    await transport.put_files([(local_tmp, remote_tmp)])
+                    await transport.run(
+                        f"mv -f {shlex.quote(remote_tmp)} {shlex.quote(remote_path)}"
+                    )
+                finally:
+                    try:
+                        os.remove(local_tmp)
+                    except OSError:
+                        pass
+            transport_pool.store_check(key, check_name, remote_path)
+            return remote_path
 
     async def _dispatch_worker(
         self,
