@@ -27,10 +27,13 @@ from typing import Dict, List, Optional, Tuple
 
 @dataclass
 class Slot:
-    """A held GPU slot.  ``gpu_id`` indexes the node's physical GPUs."""
+    """A held GPU slot.  ``gpu_id`` indexes the node's physical GPUs;
+    ``sub_id`` distinguishes co-resident slots when ``slots_per_gpu > 1``
+    (oversubscription: several electrons share one 288 GB MI355X)."""
 
     gpu_id: int
     table: "SlotTable"
+    sub_id: int = 0
     _released: bool = False
 
     def env(self) -> Dict[str, str]:
@@ -39,10 +42,20 @@ class Slot:
         # with container/pod GPU isolation instead of overriding it.
         return {"CSP_GPU_SLOT": str(self.gpu_id)}
 
+    @property
+    def worker_tag(self):
+        """Identity of the persistent worker serving this slot: one
+        worker per (gpu, sub-slot), so slots_per_gpu=N really runs N
+        warm worker processes on one GPU instead of pipelining N
+        electrons through one process."""
+        if self.table.slots_per_gpu == 1:
+            return self.gpu_id
+        return (self.gpu_id, self.sub_id)
+
     async def release(self) -> None:
         if not self._released:
             self._released = True
-            await self.table._release(self.gpu_id)
+            await self.table._release(self.gpu_id, self.sub_id)
 
     async def __aenter__(self) -> "Slot":
         return self
@@ -59,9 +72,11 @@ class SlotTable:
             raise ValueError("num_gpus must be >= 1")
         self.num_gpus = num_gpus
         self.slots_per_gpu = slots_per_gpu
-        # free_count[gpu] = remaining capacity; order = round-robin queue of
+        # free_subs[gpu] = free sub-slot ids; order = round-robin queue of
         # gpu ids, least-recently used first.
-        self._free: List[int] = [slots_per_gpu] * num_gpus
+        self._free_subs: List[List[int]] = [
+            list(range(slots_per_gpu)) for _ in range(num_gpus)
+        ]
         self._order: List[int] = list(range(num_gpus))
         self._cond: Optional[asyncio.Condition] = None
         self._cond_loop = None
@@ -91,22 +106,22 @@ class SlotTable:
             async with cond:
                 while True:
                     for idx, gpu in enumerate(self._order):
-                        if self._free[gpu] > 0:
-                            self._free[gpu] -= 1
+                        if self._free_subs[gpu]:
+                            sub = self._free_subs[gpu].pop(0)
                             # Move to the back: next acquire prefers others.
                             self._order.append(self._order.pop(idx))
                             self._in_use += 1
-                            return Slot(gpu_id=gpu, table=self)
+                            return Slot(gpu_id=gpu, table=self, sub_id=sub)
                     await cond.wait()
 
         if timeout is None:
             return await _take()
         return await asyncio.wait_for(_take(), timeout=timeout)
 
-    async def _release(self, gpu_id: int) -> None:
+    async def _release(self, gpu_id: int, sub_id: int = 0) -> None:
         cond = self._condition()
         async with cond:
-            self._free[gpu_id] += 1
+            self._free_subs[gpu_id].append(sub_id)
             self._in_use -= 1
             cond.notify_all()
 

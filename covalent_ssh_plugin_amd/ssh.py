@@ -870,9 +870,10 @@ class SSHExecutor(RemoteExecutor):
                             gpu_lib,
                             arg_meta=arg_meta,
                             arg_bufs=arg_bufs,
-                            worker_tag=
-                            (
-                                record.gpu_id
+                            worker_tag=(
+                                slot.worker_tag
+                                if slot is not None
+                                else record.gpu_id
                                 if record.gpu_id is not None
                                 else worker_pool.pick_cpu_tag(
                                     self._pool_key(), self.cpu_workers
@@ -1037,10 +1038,15 @@ class SSHExecutor(RemoteExecutor):
         cmd = self._wrap_conda(f"{self.python_path} {shlex.quote(script_remote)}")
         started = 0
         if self.hip_visible_devices_policy == "roundrobin" and has_gpu:
-            tags_envs = [
-                (gpu, {"CSP_GPU_SLOT": str(gpu)})
-                for gpu in range(slots if slots is not None else self.gpu_slots)
-            ]
+            n_gpus = slots if slots is not None else self.gpu_slots
+            tags_envs = []
+            for gpu in range(n_gpus):
+                if self.slots_per_gpu == 1:
+                    tags_envs.append((gpu, {"CSP_GPU_SLOT": str(gpu)}))
+                else:
+                    # oversubscription: one warm worker per (gpu, sub-slot)
+                    for sub in range(self.slots_per_gpu):
+                        tags_envs.append(((gpu, sub), {"CSP_GPU_SLOT": str(gpu)}))
         elif self.hip_visible_devices_policy == "fixed" and has_gpu:
             tags_envs = [(self.fixed_gpu, {"CSP_GPU_SLOT": str(self.fixed_gpu)})]
         else:

@@ -1,9 +1,18 @@
 """rocprof capture assertion: the CDNA4 warm-up/probe
-kernels must actually dispatch on the GPU, verified from a rocprofv3
-kernel trace rather than trusted from timings."""
+kernels must actually dispatch on the GPU, verified from a profiler
+kernel trace rather than trusted from timings.
 
+VERDICT r1 item 8: the round-1 driver box had no ``rocprofv3`` on PATH
+and the assertion silently skipped.  Now the test probes rocprofv3 AND
+classic rocprof, in PATH and /opt/rocm/bin, and verifies kernel names by
+raw-byte search across every artifact the profiler produced (works for
+rocprofv3 sqlite DBs, CSV/JSON outputs of either version).  It only
+skips when NO profiler binary exists anywhere, and the skip message
+records exactly what was probed.
+"""
+
+import os
 import shutil
-import sqlite3
 import subprocess
 import sys
 from pathlib import Path
@@ -14,36 +23,68 @@ pytestmark = pytest.mark.gpu
 
 REPO = Path(__file__).resolve().parent.parent
 
+_CANDIDATES = [
+    ("v3", "rocprofv3"),
+    ("v3", "/opt/rocm/bin/rocprofv3"),
+    ("v2", "rocprof"),
+    ("v2", "/opt/rocm/bin/rocprof"),
+]
+
+
+def _find_profiler():
+    probed = []
+    for version, cand in _CANDIDATES:
+        path = shutil.which(cand) if os.sep not in cand else (
+            cand if os.access(cand, os.X_OK) else None
+        )
+        probed.append(f"{cand}={'found' if path else 'absent'}")
+        if path:
+            return version, path, probed
+    return None, None, probed
+
 
 @pytest.mark.timeout(300)
 def test_probe_kernels_in_rocprof_trace(tmp_path):
-    if shutil.which("rocprofv3") is None:
-        pytest.skip("rocprofv3 not on PATH")
+    version, prof, probed = _find_profiler()
+    if prof is None:
+        pytest.skip("no rocprof binary found; probed: " + ", ".join(probed))
+
     out_dir = tmp_path / "prof"
+    out_dir.mkdir()
     code = (
         "import sys; sys.path.insert(0, %r); "
         "import torch; torch.cuda.init(); "
         "from covalent_ssh_plugin_amd.gpu import probe; probe.warmup(0, 30)"
     ) % str(REPO)
+    workload = [sys.executable, "-c", code]
+    if version == "v3":
+        argv = [prof, "--kernel-trace", "-d", str(out_dir), "--"] + workload
+    else:
+        # classic rocprof: per-kernel stats CSV next to -o
+        argv = [prof, "--stats", "-o", str(out_dir / "results.csv")] + workload
+
     proc = subprocess.run(
-        ["rocprofv3", "--kernel-trace", "-d", str(out_dir), "--", sys.executable, "-c", code],
-        cwd="/tmp",
-        env={"TMPDIR": "/tmp", **__import__("os").environ},
+        argv,
+        cwd=str(out_dir),
+        env={**os.environ, "TMPDIR": "/tmp"},
         capture_output=True,
         text=True,
         timeout=280,
     )
-    assert proc.returncode == 0, proc.stderr[-2000:]
-    dbs = list(out_dir.rglob("*_results.db"))
-    assert dbs, f"no rocprof results db under {out_dir}"
-    names = set()
-    for db in dbs:
-        con = sqlite3.connect(db)
-        for (table,) in con.execute(
-            "SELECT name FROM sqlite_master WHERE type='table' AND name LIKE 'rocpd_info_kernel_symbol%'"
-        ):
-            for (display,) in con.execute(f"SELECT display_name FROM {table}"):
-                names.add(display or "")
-    joined = " ".join(names)
-    assert "csp_mfma_spin_kernel" in joined, sorted(names)
-    assert "csp_hbm_sweep" in joined, sorted(names)
+    assert proc.returncode == 0, (proc.stdout[-1000:], proc.stderr[-2000:])
+
+    artifacts = [p for p in out_dir.rglob("*") if p.is_file()]
+    assert artifacts, f"{prof} produced no artifacts under {out_dir}"
+    found = set()
+    for p in artifacts:
+        try:
+            blob = p.read_bytes()
+        except OSError:
+            continue
+        for kernel in (b"csp_mfma_spin_kernel", b"csp_hbm_sweep"):
+            if kernel in blob:
+                found.add(kernel.decode())
+    assert found >= {"csp_mfma_spin_kernel", "csp_hbm_sweep"}, (
+        f"profiler={prof} artifacts={[str(p) for p in artifacts][:20]} "
+        f"found={sorted(found)}"
+    )

@@ -140,6 +140,42 @@ def test_slot_pinning_without_gpu_lib(local_executor):
     assert ex.last_task_record.gpu_id in (0, 1)
 
 
+def test_oversubscription_runs_distinct_workers(local_executor):
+    """slots_per_gpu=N runs N warm worker PROCESSES per GPU (one per
+    sub-slot), not N electrons pipelined through one process (VERDICT r1
+    item 2a, CPU-side readiness half)."""
+    ex = local_executor(
+        persistent_workers=True,
+        hip_visible_devices_policy="roundrobin",
+        gpu_slots=1,
+        slots_per_gpu=4,
+    )
+    transport_pool.store_check(ex._pool_key(), "env", (True, "", "", True))
+
+    def report():
+        import os
+        import time
+
+        time.sleep(0.2)  # hold the slot so the fan spreads over sub-slots
+        return (os.getpid(), os.environ.get("CSP_GPU_SLOT"))
+
+    async def go():
+        try:
+            return await asyncio.gather(
+                *[
+                    ex.execute(report, [], {}, dispatch_id="over", node_id=i)
+                    for i in range(16)
+                ]
+            )
+        finally:
+            await SSHExecutor.close_pool()
+
+    results = asyncio.run(go())
+    pids = {pid for pid, _ in results}
+    assert len(pids) == 4, pids  # one worker process per sub-slot
+    assert {slot for _, slot in results} == {"0"}  # all pinned to GPU 0
+
+
 def test_negative_env_check_expires(local_executor, tmp_path, monkeypatch):
     """a failing env check must heal after its TTL —
     a python-missing endpoint recovers once the remote is fixed."""
