@@ -105,5 +105,18 @@ class LocalTransport(Transport):
         )
         return Channel(proc, label=f"local-worker")
 
+    async def open_pipe(self, command: str, env: Optional[dict] = None):
+        full_cmd = self._env_prefix(env) + command
+        return await asyncio.create_subprocess_exec(
+            "bash",
+            "-c",
+            full_cmd,
+            cwd=str(self._home),
+            stdin=asyncio.subprocess.PIPE,
+            stdout=asyncio.subprocess.PIPE,
+            stderr=asyncio.subprocess.PIPE,
+            limit=4 * 1024 * 1024,
+        )
+
     async def close(self) -> None:
         self._connected = False

@@ -187,6 +187,16 @@ class OpenSSHTransport(Transport):
         )
         return Channel(proc, label=f"worker@{self.endpoint}")
 
+    async def open_pipe(self, command: str, env: Optional[dict] = None):
+        argv = self._base_args() + [self.hostname, "--", self._env_prefix(env) + command]
+        return await asyncio.create_subprocess_exec(
+            *argv,
+            stdin=asyncio.subprocess.PIPE,
+            stdout=asyncio.subprocess.PIPE,
+            stderr=asyncio.subprocess.PIPE,
+            limit=4 * 1024 * 1024,
+        )
+
     async def close(self) -> None:
         if not self._connected:
             return

@@ -374,6 +374,20 @@ def test_worker_tensor_over_4gib(tmp_path):
     asyncio.run(ex.close_pool())
 
 
+@pytest.mark.timeout(600)
+def test_fused_45gib_result_bounded_rss(tmp_path):
+    """VERDICT r1 item 4 done-criterion: a 4.5 GiB stub-path (fused)
+    result with peak dispatcher RSS < 1.5x the payload.  The result
+    streams through the sentinel parser straight to disk; only the
+    final deserialized object occupies dispatcher memory."""
+    from test_fused_streaming import _run_rss_probe
+
+    nbytes = (4 * 1024 + 512) << 20  # 4.5 GiB
+    peak = _run_rss_probe(tmp_path, nbytes)
+    print(f"dispatcher peak RSS: {peak / 1e9:.2f} GB for {nbytes / 1e9:.2f} GB payload")
+    assert peak < nbytes * 1.5, peak
+
+
 def test_mfma_numerics_vs_torch_reference(gpu_lib):
     """The warm-up kernel's MFMA instruction, numerics-checked: one
     v_mfma_f32_32x32x16_bf16 with self-described operand layouts must

@@ -24,18 +24,49 @@ binary = st.binary(max_size=2048).filter(
 )
 
 
+def _parse_chunked(stream: bytes, chunks):
+    """Feed ``stream`` through FusedStreamParser at the given chunk
+    boundaries; returns (task_out, result_bytes_or_None, meta_or_None)."""
+    from covalent_ssh_plugin_amd.ssh import FusedStreamParser
+
+    sink = bytearray()
+    parser = FusedStreamParser(S_RESULT, S_META, sink.extend)
+    pos = 0
+    for size in chunks:
+        parser.feed(stream[pos : pos + size])
+        pos += size
+    parser.feed(stream[pos:])
+    parser.finish()
+    result = bytes(sink) if parser.have_result else None
+    if parser.have_result:
+        import hashlib
+
+        assert parser.sha_hex == hashlib.sha256(sink).hexdigest()
+    return bytes(parser.task_out), result, parser.meta_bytes
+
+
 @settings(max_examples=200, deadline=None)
-@given(task_out=binary, result=binary, meta=binary)
-def test_split_stream_roundtrip(task_out, result, meta):
+@given(
+    task_out=binary,
+    result=binary,
+    meta=binary,
+    chunks=st.lists(st.integers(min_value=1, max_value=97), max_size=40),
+)
+def test_split_stream_roundtrip(task_out, result, meta, chunks):
+    """The streaming splitter must reproduce the exact segments under
+    ARBITRARY chunk boundaries (sentinels straddling reads included)."""
     stream = task_out + S_RESULT + result + S_META + meta
-    t, r, m = SSHExecutor._split_stream(stream, S_RESULT, S_META)
+    t, r, m = _parse_chunked(stream, chunks)
     assert (t, r, m) == (task_out, result, meta)
 
 
 @settings(max_examples=100, deadline=None)
-@given(task_out=binary)
-def test_split_stream_no_result(task_out):
-    t, r, m = SSHExecutor._split_stream(task_out, S_RESULT, S_META)
+@given(
+    task_out=binary,
+    chunks=st.lists(st.integers(min_value=1, max_value=97), max_size=40),
+)
+def test_split_stream_no_result(task_out, chunks):
+    t, r, m = _parse_chunked(task_out, chunks)
     assert t == task_out and r is None and m is None
 
 

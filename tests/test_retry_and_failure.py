@@ -42,6 +42,9 @@ class FlakyTransport(Transport):
     async def open_channel(self, command, env=None):
         raise NotImplementedError
 
+    async def open_pipe(self, command, env=None):
+        raise NotImplementedError
+
     async def close(self):
         self._connected = False
 

@@ -134,22 +134,33 @@ def test_env_prefix_quoting(tmp_path):
 
 
 def test_split_stream_sentinels():
-    from covalent_ssh_plugin_amd.ssh import SSHExecutor
+    from covalent_ssh_plugin_amd.ssh import FusedStreamParser
 
     s_result, s_meta = b"\n--R--\n", b"\n--M--\n"
     task_out = b"user output \x00\xff --R not quite"
     result = b"\x80\x04binary pickle"
     meta = b'{"phases_ms": {}}'
+
+    def parse(stream, chunk=5):
+        sink = bytearray()
+        p = FusedStreamParser(s_result, s_meta, sink.extend)
+        for i in range(0, len(stream), chunk):
+            p.feed(stream[i : i + chunk])
+        p.finish()
+        return (
+            bytes(p.task_out),
+            bytes(sink) if p.have_result else None,
+            p.meta_bytes,
+        )
+
     stream = task_out + s_result + result + s_meta + meta
-    t, r, m = SSHExecutor._split_stream(stream, s_result, s_meta)
-    assert (t, r, m) == (task_out, result, meta)
+    assert parse(stream) == (task_out, result, meta)
 
     # no sentinel at all (task failed before writing the result)
-    t, r, m = SSHExecutor._split_stream(b"just logs", s_result, s_meta)
-    assert t == b"just logs" and r is None and m is None
+    assert parse(b"just logs") == (b"just logs", None, None)
 
     # result but no meta
-    t, r, m = SSHExecutor._split_stream(task_out + s_result + result, s_result, s_meta)
+    t, r, m = parse(task_out + s_result + result)
     assert r == result and m is None
 
 
