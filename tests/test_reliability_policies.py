@@ -140,6 +140,22 @@ def test_slot_pinning_without_gpu_lib(local_executor):
     assert ex.last_task_record.gpu_id in (0, 1)
 
 
+def test_fixed_policy_pins_to_fixed_gpu(local_executor):
+    """`fixed` policy (per-rank bench drivers): every electron pins to
+    fixed_gpu with no slot accounting."""
+    ex = local_executor(hip_visible_devices_policy="fixed", fixed_gpu=3, gpu_slots=8)
+    transport_pool.store_check(ex._pool_key(), "env", (True, "", "", True))
+
+    def report():
+        import os
+
+        return (os.environ.get("CSP_GPU_SLOT"), os.environ.get("HIP_VISIBLE_DEVICES"))
+
+    slot, hip = asyncio.run(ex.execute(report, [], {}, dispatch_id="fx", node_id=0))
+    assert (slot, hip) == ("3", "3")
+    assert ex.last_task_record.gpu_id == 3
+
+
 def test_oversubscription_runs_distinct_workers(local_executor):
     """slots_per_gpu=N runs N warm worker PROCESSES per GPU (one per
     sub-slot), not N electrons pipelined through one process (VERDICT r1
