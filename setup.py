@@ -29,7 +29,13 @@ setup(
     package_data={
         "covalent_ssh_plugin_amd": ["ops/*.so", "ops/hip/*.hip", "remote/stub_template.py"],
     },
-    install_requires=["cloudpickle>=2.0"],
+    # runtime deps live in requirements.txt (reference setup.py:28-34
+    # parses the same file; license CI checks it)
+    install_requires=[
+        line.strip()
+        for line in (here / "requirements.txt").read_text().splitlines()
+        if line.strip() and not line.startswith("#")
+    ],
     extras_require={"covalent": ["covalent>=0.202.0,<1"]},
     entry_points={
         "covalent.executor.executor_plugins": [
