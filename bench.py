@@ -58,7 +58,7 @@ SSHIM_PASS_ENV = (
 def build_electron(config: str):
     """Electron bodies are defined inside a factory so cloudpickle
     serializes them by value."""
-    if config in ("noop", "noop-stub", "fan"):
+    if config in ("noop", "noop-stub", "noop-isolated", "fan"):
 
         def noop():
             return 0
@@ -110,7 +110,8 @@ def build_electron(config: str):
 def default_steps(config: str) -> int:
     # long enough to amortize ~30 ms OS hiccups for the fast configs,
     # bounded wall time for the heavy ones
-    return {"staging": 8, "rccl": 3, "fan": 64, "mm": 256}.get(config, 2048)
+    return {"staging": 8, "rccl": 3, "fan": 64, "mm": 256,
+            "noop-isolated": 512}.get(config, 2048)
 
 
 def default_warmup(config: str) -> int:
@@ -166,8 +167,11 @@ async def bench_once(
         warmup_gpu=config in ("mm", "staging"),
         pinned_staging_threshold_bytes=1 << 20,
         # warm worker per GPU slot (the production dispatch path);
-        # "noop-stub" measures the classic spawn-per-task stub
+        # "noop-stub" measures the classic spawn-per-task stub;
+        # "noop-isolated" measures fork-isolated dispatch (fresh child
+        # process per electron off a warm zygote)
         persistent_workers=config != "noop-stub",
+        isolate_tasks=config == "noop-isolated",
         **transport_kwargs,
     )
 
@@ -379,8 +383,10 @@ def main() -> None:
             "dtype": "bf16" if args.config in ("mm", "staging") else "n/a",
             "data": (
                 f"synthetic electrons, {transport_note}, "
-                + ("persistent-worker dispatch" if args.config != "noop-stub"
-                   else "spawn-per-task stub dispatch")
+                + {
+                    "noop-stub": "spawn-per-task stub dispatch",
+                    "noop-isolated": "fork-isolated dispatch (fresh child per electron)",
+                }.get(args.config, "persistent-worker dispatch")
             ),
             "p50_ms": round(p50_ms, 3),
             "p99_ms": round(p99_ms, 3),

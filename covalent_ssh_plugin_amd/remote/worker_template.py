@@ -53,6 +53,12 @@ GPU_LIB = "__CSP_GPU_LIB__"
 DO_WARMUP = bool(__CSP_WARMUP__)
 STAGING_THRESHOLD = int(__CSP_STAGING_THRESHOLD__)
 IDLE_TIMEOUT = float(__CSP_IDLE_TIMEOUT__)  # seconds; 0 = never exit
+# Fork-isolation mode: the worker is a warm ZYGOTE (python + cloudpickle
+# + torch imported, HIP **not** initialized — initializing HIP pre-fork
+# would break the children) and every electron executes in a freshly
+# forked child that does its own GPU prologue.  Fresh-process semantics
+# at fork cost instead of a full interpreter + import start per task.
+ISOLATE = bool(__CSP_ISOLATE__)
 
 if GPU_LIB:
     GPU_LIB = os.path.abspath(os.path.expanduser(GPU_LIB))
@@ -367,11 +373,101 @@ def _serve_one(request):
     return result_blob, meta, buffers
 
 
+def _serve_isolated(request):
+    """Run one electron in a freshly forked child (fresh-process
+    semantics: no state, no HIP context, no module cache survives into
+    the next electron).  The child writes its response frames to a pipe;
+    the parent relays them frame-by-frame so a child crash can never
+    leave a half-written frame on the protocol stream."""
+    global _served
+    _served += 1
+    serial = _served
+    r, w = os.pipe()
+    pid = os.fork()
+    if pid == 0:
+        # ---- child: the per-task process ----
+        os.close(r)
+        try:
+            import ctypes
+
+            libc = ctypes.CDLL(None, use_errno=True)
+            libc.prctl(1, 9, 0, 0, 0)  # PR_SET_PDEATHSIG=SIGKILL: die with parent
+        except Exception:  # noqa: BLE001
+            pass
+        try:
+            if GPU_LIB and GPU_SLOT is not None and DO_WARMUP:
+                _prologue()  # fresh HIP init belongs to THIS child
+            result_blob, meta, buffers = _serve_one(request)
+            meta["isolated"] = True
+            meta["served"] = serial
+            _write_frame(w, pickle.dumps(("R1", result_blob, meta, len(buffers))))
+            for view, _keep in buffers:
+                _write_frame(w, view)
+        except BaseException as e:  # noqa: BLE001 - report anything reportable
+            try:
+                blob = pickle.dumps(
+                    (None, RuntimeError(f"isolated task failed: {e!r}"))
+                )
+                _write_frame(
+                    w, pickle.dumps(("R1", blob, {"phases_ms": {}, "isolated": True}, 0))
+                )
+            except BaseException:  # noqa: BLE001
+                pass
+        os._exit(0)
+
+    # ---- parent: frame-by-frame relay ----
+    os.close(w)
+    relayed = 0
+    expected = None
+    while True:
+        frame = _read_frame(r)
+        if frame is None or frame == b"":
+            break
+        if relayed == 0:
+            _tag, _blob, _meta, nbuf = pickle.loads(frame)
+            expected = 1 + nbuf
+        _write_frame(_proto_fd, frame)
+        relayed += 1
+        if expected is not None and relayed >= expected:
+            break
+    os.close(r)
+    _, status = os.waitpid(pid, 0)
+    if relayed == 0:
+        # child died before completing even the main response frame:
+        # synthesize the error so the electron fails cleanly and the
+        # zygote keeps serving (no worker respawn needed)
+        blob = pickle.dumps(
+            (None, RuntimeError(
+                f"isolated task process died (wait status {status}) "
+                "before reporting a result"
+            ))
+        )
+        _write_frame(
+            _proto_fd,
+            pickle.dumps(("R1", blob, {"phases_ms": {}, "isolated": True,
+                                       "served": serial}, 0)),
+        )
+    elif expected is not None and relayed < expected:
+        # partial multi-frame response already relayed: the stream is
+        # unrecoverable — exit so the dispatcher sees ChannelClosed
+        os._exit(4)
+
+
 def main():
     startup_meta = {"pid": os.getpid(), "gpu": None, "error": None,
-                    "gpu_slot": GPU_SLOT,
+                    "gpu_slot": GPU_SLOT, "isolate": ISOLATE,
                     "hip_visible_devices": os.environ.get("HIP_VISIBLE_DEVICES")}
-    if GPU_LIB and GPU_SLOT is not None:
+    if ISOLATE:
+        # zygote warm-up: bind the expensive imports pre-fork, but touch
+        # NO GPU state (HIP contexts do not survive fork)
+        try:
+            import importlib.util
+
+            if importlib.util.find_spec("torch") is not None:
+                import torch  # noqa: F401
+        except Exception as e:  # noqa: BLE001
+            startup_meta["error"] = repr(e)
+    elif GPU_LIB and GPU_SLOT is not None:
         try:
             _prologue()
             startup_meta["gpu"] = _gpu_info
@@ -398,6 +494,9 @@ def main():
         # ack AFTER the full request is in hand, BEFORE user code runs:
         # the dispatcher's retry-on-death policy keys off this frame
         _write_frame(_proto_fd, pickle.dumps(("A1", request.get("op_id"))))
+        if ISOLATE:
+            _serve_isolated(request)
+            continue
         result_blob, meta, buffers = _serve_one(request)
         _write_frame(_proto_fd, pickle.dumps(("R1", result_blob, meta, len(buffers))))
         for view, _keep in buffers:

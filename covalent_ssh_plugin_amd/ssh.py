@@ -87,6 +87,13 @@ _EXECUTOR_PLUGIN_DEFAULTS = {
     "pinned_staging_threshold_bytes": DEFAULT_STAGING_THRESHOLD,
     "batch_roundtrips": True,  # fused single-round-trip dispatch
     "persistent_workers": False,  # warm worker process per GPU slot
+    # Fork-isolated dispatch: persistent ZYGOTE worker (python + torch
+    # imported, HIP untouched) forks a fresh child per electron —
+    # spawn-per-task's fresh-process semantics (no state/HIP/module
+    # leakage between electrons, crashes contained) at fork cost
+    # (~ms) instead of a full interpreter + torch import start.
+    # Implies worker-channel dispatch.
+    "isolate_tasks": False,
     "cpu_workers": 4,  # worker-set size when no GPU policy is active
     "task_timeout": 0,  # seconds; 0 = unlimited
     "worker_idle_timeout": 0,  # seconds; 0 = workers never exit on idle
@@ -269,6 +276,7 @@ class SSHExecutor(RemoteExecutor):
         pinned_staging_threshold_bytes: Optional[int] = None,
         batch_roundtrips: Optional[bool] = None,
         persistent_workers: Optional[bool] = None,
+        isolate_tasks: Optional[bool] = None,
         cpu_workers: Optional[int] = None,
         task_timeout: Optional[float] = None,
         worker_idle_timeout: Optional[float] = None,
@@ -314,9 +322,10 @@ class SSHExecutor(RemoteExecutor):
             _conf("pinned_staging_threshold_bytes", pinned_staging_threshold_bytes)
         )
         self.batch_roundtrips = bool(_conf("batch_roundtrips", batch_roundtrips, default=True))
+        self.isolate_tasks = bool(_conf("isolate_tasks", isolate_tasks, default=False))
         self.persistent_workers = bool(
             _conf("persistent_workers", persistent_workers, default=False)
-        )
+        ) or self.isolate_tasks  # isolation rides the worker channel
         self.cpu_workers = int(_conf("cpu_workers", cpu_workers))
         self.task_timeout = float(_conf("task_timeout", task_timeout, default=0) or 0)
         self.worker_idle_timeout = float(
@@ -755,6 +764,7 @@ class SSHExecutor(RemoteExecutor):
             warmup=self.warmup_gpu,
             staging_threshold=self.pinned_staging_threshold_bytes,
             idle_timeout=self.worker_idle_timeout,
+            isolate=self.isolate_tasks,
         )
         digest = _script_digest(text)
         key = self._pool_key()

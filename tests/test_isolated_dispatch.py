@@ -1,0 +1,179 @@
+"""Fork-isolated dispatch (isolate_tasks=True): a persistent zygote
+worker forks a fresh child per electron — spawn-per-task's process
+isolation (no state leakage, crash containment) at fork cost instead of
+a full interpreter + import start per task.
+"""
+
+import asyncio
+
+import pytest
+
+from covalent_ssh_plugin_amd import SSHExecutor
+
+
+def _iso(local_executor, **kw):
+    kw.setdefault("isolate_tasks", True)
+    kw.setdefault("cpu_workers", 1)
+    return local_executor(**kw)
+
+
+def test_each_electron_gets_fresh_process(local_executor):
+    ex = _iso(local_executor)
+
+    def pid():
+        import os
+
+        return os.getpid()
+
+    async def go():
+        try:
+            a = await ex.execute(pid, [], {}, dispatch_id="i", node_id=0)
+            meta_a = dict(ex.last_task_record.remote_meta)
+            b = await ex.execute(pid, [], {}, dispatch_id="i", node_id=1)
+            return a, meta_a, b
+        finally:
+            await SSHExecutor.close_pool()
+
+    a, meta_a, b = asyncio.run(go())
+    assert a != b, "electrons shared a process in isolate mode"
+    assert meta_a["isolated"] is True
+    assert meta_a["pid"] == a  # meta reports the child, not the zygote
+
+
+def test_no_state_leaks_between_electrons(local_executor):
+    """The semantic contract isolation buys: module-global state set by
+    one electron is INVISIBLE to the next (persistent workers by design
+    share it — asserted as the contrast)."""
+
+    def set_flag():
+        import os
+
+        os.environ["CSP_LEAKY_STATE"] = "set"
+        import builtins
+
+        builtins._csp_leak = 42
+        return "set"
+
+    def read_flag():
+        import os
+
+        import builtins
+
+        return (os.environ.get("CSP_LEAKY_STATE"), getattr(builtins, "_csp_leak", None))
+
+    async def run_pair(ex):
+        try:
+            await ex.execute(set_flag, [], {}, dispatch_id="s", node_id=0)
+            return await ex.execute(read_flag, [], {}, dispatch_id="s", node_id=1)
+        finally:
+            await SSHExecutor.close_pool()
+
+    iso = _iso(local_executor)
+    assert asyncio.run(run_pair(iso)) == (None, None)
+
+    warm = local_executor(persistent_workers=True, cpu_workers=1)
+    assert asyncio.run(run_pair(warm)) == ("set", 42)
+
+
+def test_crash_contained_no_respawn(local_executor):
+    """A hard-crashing electron (os._exit bypasses exception capture)
+    fails cleanly; the zygote keeps serving WITHOUT a worker respawn."""
+    ex = _iso(local_executor)
+
+    def hard_crash():
+        import os
+
+        os._exit(13)
+
+    def add(x, y):
+        return x + y
+
+    async def go():
+        try:
+            with pytest.raises(RuntimeError, match="died"):
+                await ex.execute(hard_crash, [], {}, dispatch_id="c", node_id=0)
+            return await ex.execute(add, [4, 5], {}, dispatch_id="c", node_id=1)
+        finally:
+            await SSHExecutor.close_pool()
+
+    assert asyncio.run(go()) == 9
+    assert ex.counters["worker_respawns"] == 0
+
+
+def test_signal_kill_contained(local_executor):
+    ex = _iso(local_executor)
+
+    def sigkill_self():
+        import os
+        import signal
+
+        os.kill(os.getpid(), signal.SIGKILL)
+
+    async def go():
+        try:
+            with pytest.raises(RuntimeError, match="died"):
+                await ex.execute(sigkill_self, [], {}, dispatch_id="k", node_id=0)
+            return await ex.execute(lambda: "alive", [], {}, dispatch_id="k", node_id=1)
+        finally:
+            await SSHExecutor.close_pool()
+
+    assert asyncio.run(go()) == "alive"
+
+
+def test_exception_roundtrip_isolated(local_executor):
+    ex = _iso(local_executor)
+
+    def boom():
+        raise ValueError("isolated failure")
+
+    async def go():
+        try:
+            with pytest.raises(ValueError, match="isolated failure"):
+                await ex.execute(boom, [], {}, dispatch_id="e", node_id=0)
+        finally:
+            await SSHExecutor.close_pool()
+
+    asyncio.run(go())
+
+
+def test_oob_tensor_result_isolated(local_executor):
+    """Out-of-band tensor frames relay through the zygote byte-exact."""
+    torch = pytest.importorskip("torch")
+
+    def make(n):
+        import torch
+
+        return torch.arange(n, dtype=torch.float32)
+
+    ex = _iso(local_executor, pinned_staging_threshold_bytes=1 << 20)
+
+    async def go():
+        try:
+            return await ex.execute(make, [1 << 20], {}, dispatch_id="t", node_id=0)
+        finally:
+            await SSHExecutor.close_pool()
+
+    out = asyncio.run(go())
+    assert out.shape == (1 << 20,)
+    assert out[-1].item() == float((1 << 20) - 1)
+
+
+def test_isolated_over_sshim(sshim_executor):
+    """Isolation mode through the real OpenSSH transport."""
+    ex = sshim_executor(isolate_tasks=True, cpu_workers=1)
+
+    def pid():
+        import os
+
+        return os.getpid()
+
+    async def go():
+        try:
+            a = await ex.execute(pid, [], {}, dispatch_id="si", node_id=0)
+            b = await ex.execute(pid, [], {}, dispatch_id="si", node_id=1)
+            return a, b
+        finally:
+            await SSHExecutor.close_pool()
+
+    a, b = asyncio.run(go())
+    assert a != b
