@@ -135,6 +135,12 @@ class SSHTaskError(RuntimeError):
     """Dispatcher-side failure of the SSH pipeline (not the user task)."""
 
 
+class SSHConnectError(SSHTaskError):
+    """The endpoint could not be reached / verified BEFORE any task code
+    could have started (connect exhaustion, environment-check failure).
+    Safe to fail over to another node — nothing executed."""
+
+
 class FusedStreamParser:
     """Incremental splitter for the fused dispatch's stdout stream:
     ``<task stdout> s_result <result bytes> s_meta <meta json>``.
@@ -397,15 +403,23 @@ class SSHExecutor(RemoteExecutor):
     # ------------------------------------------------------------------
 
     async def _on_ssh_fail(
-        self, function: Callable, args: list, kwargs: dict, message: str
+        self,
+        function: Callable,
+        args: list,
+        kwargs: dict,
+        message: str,
+        error_cls: type = SSHTaskError,
     ) -> Any:
         """If ``run_local_on_ssh_fail``, run the task on the dispatcher
-        host; otherwise raise (reference ssh.py:202-208)."""
+        host; otherwise raise (reference ssh.py:202-208; the reference
+        raises RuntimeError — SSHTaskError/SSHConnectError are
+        RuntimeError subclasses, so the surface is compatible while the
+        cluster's failover can distinguish pre-execution failures)."""
         self.counters["ssh_failures"] += 1
         app_log.warning("SSH dispatch failed: %s", message)
         if self.run_local_on_ssh_fail:
             return await asyncio.to_thread(function, *args, **kwargs)
-        raise RuntimeError(message)
+        raise error_cls(message)
 
     # ------------------------------------------------------------------
     # RemoteExecutor template methods (reference ssh.py:317-464)
@@ -1103,7 +1117,9 @@ class SSHExecutor(RemoteExecutor):
                 transport = await self._client_connect()
             except (TransportConnectError, OSError) as e:
                 return await self._on_ssh_fail(
-                    function, args, kwargs, f"Could not connect to {self.hostname}: {e}"
+                    function, args, kwargs,
+                    f"Could not connect to {self.hostname}: {e}",
+                    error_cls=SSHConnectError,
                 )
         if transport is None:
             return await self._on_ssh_fail(
@@ -1112,13 +1128,17 @@ class SSHExecutor(RemoteExecutor):
                 kwargs,
                 f"Could not connect to {self.hostname} after "
                 f"{self.max_connection_attempts} attempts",
+                error_cls=SSHConnectError,
             )
 
         with timer.phase("env_checks"):
             try:
                 gpu_lib, has_gpu = await self._ensure_environment(transport)
             except SSHTaskError as e:
-                return await self._on_ssh_fail(function, args, kwargs, str(e))
+                # pre-execution: nothing has run yet -> failover-safe
+                return await self._on_ssh_fail(
+                    function, args, kwargs, str(e), error_cls=SSHConnectError
+                )
 
         slot = None
         # Slot policy keys off GPU PRESENCE, not library provisioning:

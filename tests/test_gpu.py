@@ -311,6 +311,43 @@ def test_executor_gpu_electron_over_sshim(tmp_path, sshim, monkeypatch):
     asyncio.run(ex.close_pool())
 
 
+def test_isolated_gpu_electron(tmp_path):
+    """Fork-isolated dispatch on a real MI355X: each electron's child
+    process does its OWN HIP init and pinned staging; two electrons get
+    two distinct processes with bit-exact results."""
+
+    def electron(n):
+        import os
+
+        import torch
+
+        a = torch.randn(n, n, device="cuda", dtype=torch.bfloat16)
+        b = torch.eye(n, device="cuda", dtype=torch.bfloat16)
+        c = a @ b
+        torch.cuda.synchronize()
+        return {"pid": os.getpid(), "mat": c, "check": a}
+
+    ex = _executor(
+        tmp_path, isolate_tasks=True, pinned_staging_threshold_bytes=1024
+    )
+
+    async def go():
+        try:
+            r1 = await ex.execute(electron, [1024], {}, dispatch_id="iso", node_id=0)
+            m1 = dict(ex.last_task_record.remote_meta)
+            r2 = await ex.execute(electron, [1024], {}, dispatch_id="iso", node_id=1)
+            return r1, m1, r2
+        finally:
+            await ex.close_pool()
+
+    r1, m1, r2 = asyncio.run(go())
+    assert torch.equal(r1["mat"], r1["check"])
+    assert torch.equal(r2["mat"], r2["check"])
+    assert r1["pid"] != r2["pid"], "isolated electrons shared a process"
+    assert m1["isolated"] is True
+    assert m1["staging"]["mode"] == "pinned", m1
+
+
 def test_probe_props_fast(gpu_lib):
     import time
 
