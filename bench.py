@@ -172,6 +172,10 @@ async def bench_once(
         # process per electron off a warm zygote)
         persistent_workers=config != "noop-stub",
         isolate_tasks=config == "noop-isolated",
+        # noop electrons don't need torch in the children; "none" makes
+        # forks millisecond-cheap (preloading torch-ROCm costs ~250 ms
+        # of address-space copying per fork on MI355X)
+        isolate_preload="none" if config == "noop-isolated" else "",
         **transport_kwargs,
     )
 

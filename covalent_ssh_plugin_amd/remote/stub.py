@@ -61,11 +61,13 @@ def render_worker(
     staging_threshold: Optional[int] = None,
     idle_timeout: float = 0.0,
     isolate: bool = False,
+    isolate_preload: str = "torch",
 ) -> str:
     """Return the persistent worker script text (one per endpoint; the
     GPU slot arrives via the CSP_GPU_SLOT env var at launch).  With
     ``isolate`` the worker is a fork-server zygote: every electron runs
-    in a freshly forked child process."""
+    in a freshly forked child process; ``isolate_preload`` picks what
+    the zygote imports pre-fork ("torch" or "none")."""
     text = _template_text(str(_WORKER_TEMPLATE_PATH))
     thr = DEFAULT_STAGING_THRESHOLD if staging_threshold is None else int(staging_threshold)
     replacements = {
@@ -74,6 +76,7 @@ def render_worker(
         "__CSP_STAGING_THRESHOLD__": str(thr),
         "__CSP_IDLE_TIMEOUT__": repr(float(idle_timeout)),
         "__CSP_ISOLATE__": repr(bool(isolate)),
+        "__CSP_ISOLATE_PRELOAD__": str(isolate_preload),
     }
     for token, value in replacements.items():
         text = text.replace(token, value)

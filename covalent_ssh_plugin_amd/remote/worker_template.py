@@ -54,11 +54,17 @@ DO_WARMUP = bool(__CSP_WARMUP__)
 STAGING_THRESHOLD = int(__CSP_STAGING_THRESHOLD__)
 IDLE_TIMEOUT = float(__CSP_IDLE_TIMEOUT__)  # seconds; 0 = never exit
 # Fork-isolation mode: the worker is a warm ZYGOTE (python + cloudpickle
-# + torch imported, HIP **not** initialized — initializing HIP pre-fork
-# would break the children) and every electron executes in a freshly
-# forked child that does its own GPU prologue.  Fresh-process semantics
-# at fork cost instead of a full interpreter + import start per task.
+# imported, HIP **not** initialized — initializing HIP pre-fork would
+# break the children) and every electron executes in a freshly forked
+# child that does its own GPU prologue.  Fresh-process semantics at fork
+# cost instead of a full interpreter + import start per task.
 ISOLATE = bool(__CSP_ISOLATE__)
+# What the zygote binds before forking.  "torch": children inherit a
+# loaded torch — saves the multi-second import for torch-using electrons
+# but makes each fork copy the torch-ROCm address space (measured
+# ~250 ms/fork on MI355X vs ~ms without).  "none": cheap forks; torch
+# electrons pay their own import.  Choose per workload.
+ISOLATE_PRELOAD = "__CSP_ISOLATE_PRELOAD__"
 
 if GPU_LIB:
     GPU_LIB = os.path.abspath(os.path.expanduser(GPU_LIB))
@@ -458,15 +464,16 @@ def main():
                     "gpu_slot": GPU_SLOT, "isolate": ISOLATE,
                     "hip_visible_devices": os.environ.get("HIP_VISIBLE_DEVICES")}
     if ISOLATE:
-        # zygote warm-up: bind the expensive imports pre-fork, but touch
-        # NO GPU state (HIP contexts do not survive fork)
-        try:
-            import importlib.util
+        # zygote warm-up: bind the configured imports pre-fork, but
+        # touch NO GPU state (HIP contexts do not survive fork)
+        if ISOLATE_PRELOAD == "torch":
+            try:
+                import importlib.util
 
-            if importlib.util.find_spec("torch") is not None:
-                import torch  # noqa: F401
-        except Exception as e:  # noqa: BLE001
-            startup_meta["error"] = repr(e)
+                if importlib.util.find_spec("torch") is not None:
+                    import torch  # noqa: F401
+            except Exception as e:  # noqa: BLE001
+                startup_meta["error"] = repr(e)
     elif GPU_LIB and GPU_SLOT is not None:
         try:
             _prologue()

@@ -94,6 +94,11 @@ _EXECUTOR_PLUGIN_DEFAULTS = {
     # (~ms) instead of a full interpreter + torch import start.
     # Implies worker-channel dispatch.
     "isolate_tasks": False,
+    # Zygote preload for isolate mode: "torch" (children inherit a
+    # loaded torch — saves multi-second imports for torch electrons, but
+    # each fork copies the torch-ROCm address space, ~250 ms on MI355X)
+    # or "none" (millisecond forks; torch electrons import themselves).
+    "isolate_preload": "torch",
     "cpu_workers": 4,  # worker-set size when no GPU policy is active
     "task_timeout": 0,  # seconds; 0 = unlimited
     "worker_idle_timeout": 0,  # seconds; 0 = workers never exit on idle
@@ -283,6 +288,7 @@ class SSHExecutor(RemoteExecutor):
         batch_roundtrips: Optional[bool] = None,
         persistent_workers: Optional[bool] = None,
         isolate_tasks: Optional[bool] = None,
+        isolate_preload: str = "",
         cpu_workers: Optional[int] = None,
         task_timeout: Optional[float] = None,
         worker_idle_timeout: Optional[float] = None,
@@ -329,6 +335,7 @@ class SSHExecutor(RemoteExecutor):
         )
         self.batch_roundtrips = bool(_conf("batch_roundtrips", batch_roundtrips, default=True))
         self.isolate_tasks = bool(_conf("isolate_tasks", isolate_tasks, default=False))
+        self.isolate_preload = _conf("isolate_preload", isolate_preload)
         self.persistent_workers = bool(
             _conf("persistent_workers", persistent_workers, default=False)
         ) or self.isolate_tasks  # isolation rides the worker channel
@@ -779,6 +786,7 @@ class SSHExecutor(RemoteExecutor):
             staging_threshold=self.pinned_staging_threshold_bytes,
             idle_timeout=self.worker_idle_timeout,
             isolate=self.isolate_tasks,
+            isolate_preload=self.isolate_preload,
         )
         digest = _script_digest(text)
         key = self._pool_key()

@@ -54,6 +54,8 @@ def main() -> None:
     ap.add_argument("--mix", action="store_true",
                     help="mixed workload: no-ops, bf16 matmuls, tensor returns, "
                          "exceptions (GPU electrons when cuda is visible)")
+    ap.add_argument("--isolate", action="store_true",
+                    help="fork-isolated dispatch: fresh child per electron")
     args = ap.parse_args()
 
     if args.mix:
@@ -103,6 +105,7 @@ def main() -> None:
                 cache_dir=cache,
                 python_path=sys.executable,
                 persistent_workers=True,
+                isolate_tasks=args.isolate,
                 cpu_workers=args.concurrency,
                 pinned_staging_threshold_bytes=1 << 20,
                 warmup_gpu=False,
