@@ -370,6 +370,7 @@ def main() -> None:
     lat_sorted = sorted(timed["latencies"])
     p50_ms = statistics.median(lat_sorted) * 1000.0
     p99_ms = lat_sorted[min(len(lat_sorted) - 1, int(len(lat_sorted) * 0.99))] * 1000.0
+    p999_ms = lat_sorted[min(len(lat_sorted) - 1, int(len(lat_sorted) * 0.999))] * 1000.0
     ms_per_step = elapsed / args.steps * 1000.0
 
     if rank == 0:
@@ -394,6 +395,7 @@ def main() -> None:
             ),
             "p50_ms": round(p50_ms, 3),
             "p99_ms": round(p99_ms, 3),
+            "p999_ms": round(p999_ms, 3),
             "phase_mean_ms": {k: round(v, 3) for k, v in timed.get("phases", {}).items()},
             "secondary": out.get("secondary", []),
             "config": {

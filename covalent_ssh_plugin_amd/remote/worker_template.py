@@ -275,8 +275,13 @@ def _read_frame(fd, idle_timeout=0.0):
 
 
 def _write_frame(fd, payload):
-    os.write(fd, struct.pack(">Q", len(payload)))
     view = memoryview(payload)
+    if view.nbytes <= 64 * 1024:
+        # small frames: single write (header + payload) — halves the
+        # per-frame syscalls on the no-op hot path
+        os.write(fd, struct.pack(">Q", view.nbytes) + bytes(view))
+        return
+    os.write(fd, struct.pack(">Q", view.nbytes))
     while view:
         written = os.write(fd, view[: 1 << 20])
         view = view[written:]
