@@ -196,3 +196,55 @@ def test_remote_python_missing_over_ssh(sshim_executor):
     ex = sshim_executor(python_path="/nonexistent/python3")
     with pytest.raises(RuntimeError):
         asyncio.run(ex.execute(_add, [1, 2], {}))
+
+
+def _with_fake_conda(monkeypatch):
+    import os
+    from pathlib import Path
+
+    fakebin = Path(__file__).parent / "sshim" / "fakebin"
+    monkeypatch.setenv(
+        "SSHIM_REMOTE_PATH", f"{fakebin}{os.pathsep}{os.environ.get('PATH', '')}"
+    )
+
+
+def test_conda_env_activation_end_to_end(sshim_executor, monkeypatch):
+    """The reference's conda wrapper (reference ssh.py:379-380) through
+    the real client: hook eval'd, env verified, task runs activated."""
+    _with_fake_conda(monkeypatch)
+    ex = sshim_executor(conda_env="ml")
+
+    def which_env():
+        import os
+
+        return os.environ.get("CONDA_DEFAULT_ENV")
+
+    assert asyncio.run(ex.execute(which_env, [], {})) == "ml"
+
+
+def test_conda_env_missing_fails(sshim_executor, monkeypatch):
+    """A conda_env absent from `conda env list` fails the env check
+    (reference ssh.py:508-519 behavior)."""
+    _with_fake_conda(monkeypatch)
+    ex = sshim_executor(conda_env="nonexistent-env")
+    with pytest.raises(RuntimeError, match="not found"):
+        asyncio.run(ex.execute(_add, [1, 2], {}))
+
+
+def test_conda_env_worker_mode(sshim_executor, monkeypatch):
+    """Persistent workers launch inside the activated env too."""
+    _with_fake_conda(monkeypatch)
+    ex = sshim_executor(conda_env="ml", persistent_workers=True)
+
+    def which_env():
+        import os
+
+        return os.environ.get("CONDA_DEFAULT_ENV")
+
+    async def go():
+        try:
+            return await ex.execute(which_env, [], {})
+        finally:
+            await SSHExecutor.close_pool()
+
+    assert asyncio.run(go()) == "ml"
