@@ -293,11 +293,10 @@ def main() -> None:
     )
 
     async def run_all() -> dict:
-        barrier = None  # timed region brackets below use barrier_sync()
         out = {}
         # warmup (untimed), then the timed region, bracketed by
         # barrier+synchronize on both sides
-        main_res = await bench_once(
+        await bench_once(  # warmup pass against the shared pools
             config=args.config, steps=0, warmup=args.warmup, fan=args.fan,
             **common,
         )

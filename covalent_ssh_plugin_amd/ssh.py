@@ -29,6 +29,7 @@ function/result pickle contract (reference ssh.py:147-150, exec.py:44-46)
 from __future__ import annotations
 
 import asyncio
+import functools
 import os
 import pickle as stdlib_pickle
 import shlex
@@ -124,9 +125,6 @@ def _conf(key: str, explicit: Any, default: Any = None) -> Any:
     if value is None or value == "":
         return default if default is not None else _EXECUTOR_PLUGIN_DEFAULTS.get(key)
     return value
-
-
-import functools
 
 
 @functools.lru_cache(maxsize=64)
@@ -1370,7 +1368,6 @@ class SSHExecutor(RemoteExecutor):
             return 0
         script_remote = await self._provision_worker_script(transport, gpu_lib)
         cmd = self._wrap_conda(f"{self.python_path} {shlex.quote(script_remote)}")
-        started = 0
         if self.hip_visible_devices_policy == "roundrobin" and has_gpu:
             n_gpus = slots if slots is not None else self.gpu_slots
             tags_envs = []
