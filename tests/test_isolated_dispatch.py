@@ -177,3 +177,58 @@ def test_isolated_over_sshim(sshim_executor):
 
     a, b = asyncio.run(go())
     assert a != b
+
+
+def test_oob_tensor_argument_isolated(local_executor):
+    """Out-of-band tensor ARGUMENTS reach the forked child intact (the
+    parent reads the raw frames, the child rebuilds the tensors)."""
+    torch = pytest.importorskip("torch")
+
+    def total(t, scale=2.0):
+        return float(t.sum().item()) * scale
+
+    big = torch.ones(1 << 19, dtype=torch.float32)  # 2 MiB, above threshold
+    ex = _iso(local_executor, pinned_staging_threshold_bytes=1 << 20)
+
+    async def go():
+        try:
+            return await ex.execute(total, [big], {"scale": 3.0})
+        finally:
+            await SSHExecutor.close_pool()
+
+    assert asyncio.run(go()) == float(1 << 19) * 3.0
+
+
+def test_isolated_zygotes_per_gpu_slot(local_executor):
+    """Roundrobin slot policy + isolation: one zygote per GPU slot, each
+    child pinned to its slot's device."""
+    from covalent_ssh_plugin_amd.transport import pool as transport_pool
+
+    ex = _iso(
+        local_executor,
+        hip_visible_devices_policy="roundrobin",
+        gpu_slots=2,
+    )
+    transport_pool.store_check(ex._pool_key(), "env", (True, "", "", True))
+
+    def report():
+        import os
+
+        return (os.getpid(), os.environ.get("HIP_VISIBLE_DEVICES"))
+
+    async def go():
+        try:
+            out = []
+            for i in range(4):
+                out.append(
+                    await ex.execute(report, [], {}, dispatch_id="zg", node_id=i)
+                )
+            return out
+        finally:
+            await SSHExecutor.close_pool()
+
+    results = asyncio.run(go())
+    # every electron ran in a distinct child process
+    assert len({pid for pid, _ in results}) == 4
+    # both GPU slots were used, and pinning reached the children
+    assert {dev for _, dev in results} == {"0", "1"}
