@@ -164,15 +164,47 @@ class OpenSSHTransport(Transport):
             )
 
     async def get_file(self, remote_path: str, local_path: str) -> None:
+        """Stream the remote file to disk in chunks — a multi-GiB result
+        fetched over the discrete template path must not hold a full
+        in-RAM copy (matches the fused path's streaming)."""
         import shlex
 
-        result = await self.run(f"cat -- {shlex.quote(remote_path)}")
-        if not result.ok:
-            raise FileNotFoundError(
-                f"{self.endpoint}:{remote_path}: {result.text_err().strip()}"
-            )
         Path(local_path).parent.mkdir(parents=True, exist_ok=True)
-        Path(local_path).write_bytes(result.stdout)
+        proc = await self.open_pipe(f"cat -- {shlex.quote(remote_path)}")
+        proc.stdin.close()
+        stderr_acc = bytearray()
+
+        async def drain_stderr():
+            while True:
+                chunk = await proc.stderr.read(1 << 20)
+                if not chunk:
+                    break
+                if len(stderr_acc) < (1 << 16):
+                    stderr_acc.extend(chunk)
+
+        err_task = asyncio.ensure_future(drain_stderr())
+        try:
+            with open(local_path, "wb") as f:
+                while True:
+                    chunk = await proc.stdout.read(1 << 20)
+                    if not chunk:
+                        break
+                    await asyncio.to_thread(f.write, chunk)
+        finally:
+            await err_task
+        rc = await proc.wait()
+        if rc != 0:
+            try:
+                os.remove(local_path)
+            except OSError:
+                pass
+            err = bytes(stderr_acc).decode(errors="replace").strip()
+            if rc == _SSH_CLIENT_ERR:
+                self._connected = False
+                raise TransportConnectError(
+                    f"ssh channel to {self.endpoint} failed: {err}"
+                )
+            raise FileNotFoundError(f"{self.endpoint}:{remote_path}: {err}")
 
     async def open_channel(self, command: str, env: Optional[dict] = None):
         from .channel import Channel
