@@ -46,10 +46,11 @@ def big(n):
 async def main():
     n = {nbytes}
     ex = SSHExecutor(
-        transport="local",
-        local_home={home!r},
         cache_dir={cache!r},
         python_path=sys.executable,
+        batch_roundtrips={batch!r},
+        poll_freq=1,
+        **{transport_kwargs!r},
     )
     print("stage:start", peak_kib(), file=sys.stderr)
     out = await ex.execute(big, [n], {{}}, dispatch_id="rss", node_id=0)
@@ -64,19 +65,33 @@ asyncio.run(main())
 """
 
 
-def _run_rss_probe(tmp_path, nbytes: int) -> int:
+def _run_rss_probe(tmp_path, nbytes: int, batch: bool = True, ssh_shim=False) -> int:
     """Dispatch an nbytes-result electron in a FRESH dispatcher process
     and return that process's peak RSS in bytes."""
     home = tmp_path / "home"
     home.mkdir(exist_ok=True)
+    env = dict(os.environ)
+    if ssh_shim:
+        key = tmp_path / "rss_key"
+        key.write_text("fake\n")
+        env["PATH"] = f"{Path(REPO) / 'tests' / 'sshim'}{os.pathsep}{env.get('PATH', '')}"
+        env["SSHIM_HOME"] = str(home)
+        transport_kwargs = dict(
+            transport="ssh", hostname="rss-node.sshim", username="u",
+            ssh_key_file=str(key),
+        )
+    else:
+        transport_kwargs = dict(transport="local", local_home=str(home))
     script = _RSS_SCRIPT.format(
-        repo=REPO, nbytes=nbytes, home=str(home), cache=str(tmp_path / "cache")
+        repo=REPO, nbytes=nbytes, cache=str(tmp_path / "cache"),
+        batch=batch, transport_kwargs=transport_kwargs,
     )
     proc = subprocess.run(
         [sys.executable, "-c", script],
         capture_output=True,
         text=True,
         timeout=300,
+        env=env,
     )
     assert proc.returncode == 0, proc.stderr[-3000:]
     print(proc.stderr[-500:])
@@ -113,6 +128,17 @@ def _group_live_members(pgid: int):
         if int(pgrp) == pgid and state not in ("Z", "X"):
             alive.append(stat.split("/")[2])
     return alive
+
+
+def test_template_path_large_result_bounded_rss_over_ssh(tmp_path):
+    """The discrete template path over the REAL client: query_result's
+    get_file streams the 384 MiB result to disk chunkwise instead of
+    buffering it through CompletedCommand."""
+    nbytes = 384 << 20
+    peak = _run_rss_probe(tmp_path, nbytes, batch=False, ssh_shim=True)
+    print(f"template-path dispatcher peak RSS: {peak / 1e6:.0f} MB "
+          f"for {nbytes / 1e6:.0f} MB payload")
+    assert peak < nbytes * 1.5, peak
 
 
 def _sleeper_factory():
