@@ -56,6 +56,8 @@ def main() -> None:
                          "exceptions (GPU electrons when cuda is visible)")
     ap.add_argument("--isolate", action="store_true",
                     help="fork-isolated dispatch: fresh child per electron")
+    ap.add_argument("--isolate-preload", default="none",
+                    help="zygote preload for --isolate: torch|none")
     args = ap.parse_args()
 
     if args.mix:
@@ -106,6 +108,7 @@ def main() -> None:
                 python_path=sys.executable,
                 persistent_workers=True,
                 isolate_tasks=args.isolate,
+                isolate_preload=args.isolate_preload,
                 cpu_workers=args.concurrency,
                 pinned_staging_threshold_bytes=1 << 20,
                 warmup_gpu=False,
