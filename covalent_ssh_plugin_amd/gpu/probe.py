@@ -80,6 +80,12 @@ def load(path: str = "") -> ctypes.CDLL:
     lib.csp_memcpy_d2h.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_size_t]
     lib.csp_memcpy_h2d.restype = ctypes.c_int
     lib.csp_memcpy_h2d.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_size_t]
+    lib.csp_mem_info.restype = ctypes.c_int
+    lib.csp_mem_info.argtypes = [
+        ctypes.c_int,
+        ctypes.POINTER(ctypes.c_double),
+        ctypes.POINTER(ctypes.c_double),
+    ]
     lib.csp_last_error.restype = ctypes.c_char_p
 
     _lib = lib
@@ -116,6 +122,16 @@ def probe_props(device: int = 0) -> dict:
     buf = ctypes.create_string_buffer(8192)
     _check(lib, lib.csp_probe_props_json(device, buf, len(buf)), "csp_probe_props_json")
     return json.loads(buf.value.decode())
+
+
+def mem_info(device: int = 0) -> dict:
+    """HBM occupancy via csp_mem_info (the per-task telemetry call)."""
+    lib = load()
+    free_gb = ctypes.c_double()
+    total_gb = ctypes.c_double()
+    _check(lib, lib.csp_mem_info(device, ctypes.byref(free_gb), ctypes.byref(total_gb)),
+           "csp_mem_info")
+    return {"hbm_free_gb": free_gb.value, "hbm_total_gb": total_gb.value}
 
 
 def warmup(device: int = 0, budget_ms: int = 50) -> None:

@@ -243,6 +243,17 @@ extern "C" int csp_probe_props_json(int device, char* buf, size_t buflen) {
     return 0;
 }
 
+// Lightweight per-task telemetry (worker meta): HBM occupancy without
+// the full props query.  One hipMemGetInfo round trip.
+extern "C" int csp_mem_info(int device, double* free_gb, double* total_gb) {
+    HIP_TRY(hipSetDevice(device));
+    size_t mem_free = 0, mem_total = 0;
+    HIP_TRY(hipMemGetInfo(&mem_free, &mem_total));
+    if (free_gb) *free_gb = (double)mem_free / 1.0e9;
+    if (total_gb) *total_gb = (double)mem_total / 1.0e9;
+    return 0;
+}
+
 extern "C" int csp_probe_json(int device, char* buf, size_t buflen) {
     HIP_TRY(hipSetDevice(device));
     hipDeviceProp_t props;

@@ -62,6 +62,7 @@ def render_worker(
     idle_timeout: float = 0.0,
     isolate: bool = False,
     isolate_preload: str = "torch",
+    telemetry_every: int = 0,
 ) -> str:
     """Return the persistent worker script text (one per endpoint; the
     GPU slot arrives via the CSP_GPU_SLOT env var at launch).  With
@@ -77,6 +78,7 @@ def render_worker(
         "__CSP_IDLE_TIMEOUT__": repr(float(idle_timeout)),
         "__CSP_ISOLATE__": repr(bool(isolate)),
         "__CSP_ISOLATE_PRELOAD__": str(isolate_preload),
+        "__CSP_TELEMETRY_EVERY__": str(int(telemetry_every)),
     }
     for token, value in replacements.items():
         text = text.replace(token, value)

@@ -65,6 +65,12 @@ ISOLATE = bool(__CSP_ISOLATE__)
 # ~250 ms/fork on MI355X vs ~ms without).  "none": cheap forks; torch
 # electrons pay their own import.  Choose per workload.
 ISOLATE_PRELOAD = "__CSP_ISOLATE_PRELOAD__"
+# Per-task GPU telemetry: sample HBM occupancy (hipMemGetInfo) every Nth
+# electron into the task meta.  0 = off (the hot no-op path stays
+# syscall-minimal); enable for GPU-heavy workloads where ~0.1 ms of
+# sampling is noise.
+TELEMETRY_EVERY = int(__CSP_TELEMETRY_EVERY__)
+_last_mem = None
 
 if GPU_LIB:
     GPU_LIB = os.path.abspath(os.path.expanduser(GPU_LIB))
@@ -147,8 +153,32 @@ def _load_gpu_lib():
     lib.csp_staging_release_all.restype = ctypes.c_int
     lib.csp_memcpy_d2h.restype = ctypes.c_int
     lib.csp_memcpy_d2h.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_size_t]
+    lib.csp_mem_info.restype = ctypes.c_int
+    lib.csp_mem_info.argtypes = [
+        ctypes.c_int,
+        ctypes.POINTER(ctypes.c_double),
+        ctypes.POINTER(ctypes.c_double),
+    ]
     lib.csp_last_error.restype = ctypes.c_char_p
     return lib
+
+
+def _sample_hbm():
+    """Cheap HBM occupancy sample for the task meta (telemetry row in
+    SURVEY.md §5; sampled every TELEMETRY_EVERY electrons)."""
+    global _last_mem
+    if _gpu_lib is None:
+        return
+    import ctypes
+
+    free_gb = ctypes.c_double()
+    total_gb = ctypes.c_double()
+    if _gpu_lib.csp_mem_info(0, ctypes.byref(free_gb), ctypes.byref(total_gb)) == 0:
+        _last_mem = {
+            "hbm_free_gb": round(free_gb.value, 2),
+            "hbm_total_gb": round(total_gb.value, 2),
+            "sampled_at_serial": _served,
+        }
 
 
 def _prologue():
@@ -314,8 +344,11 @@ def _rebuild_arg_tensors(obj, buffers, meta_list):
 def _serve_one(request):
     global _served
     _served += 1
+    if TELEMETRY_EVERY > 0 and _served % TELEMETRY_EVERY == 1 % TELEMETRY_EVERY:
+        _sample_hbm()
     t0 = time.monotonic()
     meta = {"phases_ms": {}, "gpu": _gpu_info, "staging": None,
+            "hbm": _last_mem,
             "gpu_slot": GPU_SLOT, "buffers": [],
             "hip_visible_devices": os.environ.get("HIP_VISIBLE_DEVICES"),
             "pid": os.getpid(), "worker": True, "served": _served}

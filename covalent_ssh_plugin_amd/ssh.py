@@ -101,6 +101,9 @@ _EXECUTOR_PLUGIN_DEFAULTS = {
     # or "none" (millisecond forks; torch electrons import themselves).
     "isolate_preload": "torch",
     "cpu_workers": 4,  # worker-set size when no GPU policy is active
+    # Sample HBM occupancy (hipMemGetInfo) into the task meta every Nth
+    # electron per worker; 0 = off (keeps the no-op hot path minimal).
+    "gpu_telemetry_every": 0,
     "task_timeout": 0,  # seconds; 0 = unlimited
     "worker_idle_timeout": 0,  # seconds; 0 = workers never exit on idle
     # Opt-in: re-run a task whose worker died AFTER execution had started
@@ -288,6 +291,7 @@ class SSHExecutor(RemoteExecutor):
         isolate_tasks: Optional[bool] = None,
         isolate_preload: str = "",
         cpu_workers: Optional[int] = None,
+        gpu_telemetry_every: Optional[int] = None,
         task_timeout: Optional[float] = None,
         worker_idle_timeout: Optional[float] = None,
         retry_on_worker_death: Optional[bool] = None,
@@ -338,6 +342,9 @@ class SSHExecutor(RemoteExecutor):
             _conf("persistent_workers", persistent_workers, default=False)
         ) or self.isolate_tasks  # isolation rides the worker channel
         self.cpu_workers = int(_conf("cpu_workers", cpu_workers))
+        self.gpu_telemetry_every = int(
+            _conf("gpu_telemetry_every", gpu_telemetry_every, default=0) or 0
+        )
         self.task_timeout = float(_conf("task_timeout", task_timeout, default=0) or 0)
         self.worker_idle_timeout = float(
             _conf("worker_idle_timeout", worker_idle_timeout, default=0) or 0
@@ -785,6 +792,7 @@ class SSHExecutor(RemoteExecutor):
             idle_timeout=self.worker_idle_timeout,
             isolate=self.isolate_tasks,
             isolate_preload=self.isolate_preload,
+            telemetry_every=self.gpu_telemetry_every,
         )
         digest = _script_digest(text)
         key = self._pool_key()

@@ -311,6 +311,50 @@ def test_executor_gpu_electron_over_sshim(tmp_path, sshim, monkeypatch):
     asyncio.run(ex.close_pool())
 
 
+def test_mem_info_telemetry(gpu_lib):
+    from covalent_ssh_plugin_amd.gpu import probe
+
+    info = probe.mem_info(0)
+    assert 200 < info["hbm_total_gb"] < 300  # 288 GB HBM3E
+    assert 0 < info["hbm_free_gb"] <= info["hbm_total_gb"]
+
+
+def test_worker_meta_hbm_telemetry(tmp_path):
+    """gpu_telemetry_every=1: every electron's meta carries a fresh HBM
+    occupancy sample, and allocations made by the task show up."""
+
+    def hold_memory(gib):
+        import torch
+
+        t = torch.empty(gib << 30, dtype=torch.uint8, device="cuda")
+        torch.cuda.synchronize()
+        return t.numel()
+
+    ex = _executor(
+        tmp_path, persistent_workers=True, gpu_telemetry_every=1,
+        warmup_gpu=True,
+    )
+
+    async def go():
+        try:
+            n1 = await ex.execute(hold_memory, [1], {}, dispatch_id="tm", node_id=0)
+            hbm1 = dict(ex.last_task_record.remote_meta["hbm"])
+            n2 = await ex.execute(hold_memory, [8], {}, dispatch_id="tm", node_id=1)
+            hbm2 = dict(ex.last_task_record.remote_meta["hbm"])
+            return n1, hbm1, n2, hbm2
+        finally:
+            await ex.close_pool()
+
+    n1, hbm1, n2, hbm2 = asyncio.run(go())
+    assert n1 == 1 << 30 and n2 == 8 << 30
+    assert hbm1["hbm_total_gb"] > 200
+    assert 0 < hbm1["hbm_free_gb"] <= hbm1["hbm_total_gb"]
+    # the second sample sees the first task's cached 1 GiB allocation
+    # (torch caching allocator keeps it) -> free memory dropped
+    assert hbm2["hbm_free_gb"] < hbm1["hbm_free_gb"]
+    assert hbm2["sampled_at_serial"] == hbm1["sampled_at_serial"] + 1
+
+
 def test_isolated_gpu_electron(tmp_path):
     """Fork-isolated dispatch on a real MI355X: each electron's child
     process does its OWN HIP init and pinned staging; two electrons get
