@@ -160,7 +160,7 @@ async def bench_once(
         hip_visible_devices_policy=(
             ("roundrobin" if config == "fan" else "fixed") if has_cuda else "none"
         ),
-        fixed_gpu=local_rank,
+        fixed_gpu=local_rank % max(1, gpu_count) if has_cuda else local_rank,
         gpu_slots=max(1, gpu_count) if has_cuda else 1,
         # dispatch-throughput metric: slot pinning yes, per-task clock
         # warm-up no (measured separately by --config mm)
@@ -259,7 +259,9 @@ def main() -> None:
         backend = "nccl" if has_cuda and world_size <= gpu_count else "gloo"
         dist.init_process_group(backend=backend)
         if has_cuda:
-            torch.cuda.set_device(local_rank)
+            # modulo: degenerate topologies (more ranks than GPUs, gloo
+            # fallback) must not address nonexistent devices
+            torch.cuda.set_device(local_rank % max(1, gpu_count))
 
     def barrier_sync():
         if distributed:
