@@ -12,6 +12,9 @@ reported as a clean task error without losing the worker.
 import asyncio
 import sys
 import tempfile
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 from covalent_ssh_plugin_amd import SSHExecutor
 
