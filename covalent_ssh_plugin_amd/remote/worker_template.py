@@ -438,6 +438,7 @@ def _serve_isolated(request):
             libc.prctl(1, 9, 0, 0, 0)  # PR_SET_PDEATHSIG=SIGKILL: die with parent
         except Exception:  # noqa: BLE001
             pass
+        wrote_reply = False
         try:
             if GPU_LIB and GPU_SLOT is not None and DO_WARMUP:
                 _prologue()  # fresh HIP init belongs to THIS child
@@ -445,18 +446,27 @@ def _serve_isolated(request):
             meta["isolated"] = True
             meta["served"] = serial
             _write_frame(w, pickle.dumps(("R1", result_blob, meta, len(buffers))))
+            wrote_reply = True
             for view, _keep in buffers:
                 _write_frame(w, view)
         except BaseException as e:  # noqa: BLE001 - report anything reportable
-            try:
-                blob = pickle.dumps(
-                    (None, RuntimeError(f"isolated task failed: {e!r}"))
-                )
-                _write_frame(
-                    w, pickle.dumps(("R1", blob, {"phases_ms": {}, "isolated": True}, 0))
-                )
-            except BaseException:  # noqa: BLE001
-                pass
+            # Fallback error reply ONLY if the real R1 never went out: a
+            # failure mid-buffer-stream must NOT append a second R1 that
+            # the parent would miscount as a raw buffer frame (the parent
+            # detects the short stream and drops the channel instead).
+            if not wrote_reply:
+                try:
+                    blob = pickle.dumps(
+                        (None, RuntimeError(f"isolated task failed: {e!r}"))
+                    )
+                    _write_frame(
+                        w,
+                        pickle.dumps(
+                            ("R1", blob, {"phases_ms": {}, "isolated": True}, 0)
+                        ),
+                    )
+                except BaseException:  # noqa: BLE001
+                    pass
         os._exit(0)
 
     # ---- parent: frame-by-frame relay ----
