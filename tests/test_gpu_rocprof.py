@@ -44,6 +44,66 @@ def _find_profiler():
 
 
 @pytest.mark.timeout(300)
+def test_pinned_staging_copy_in_trace(tmp_path):
+    """SURVEY.md §4.4 also asks for a capture assertion on the
+    pinned-staging path: a D2H copy of exactly our (distinctive) size
+    must appear in a rocprofv3 memory-copy trace of csp_memcpy_d2h.
+    (kernel+memory-copy tracing only — never combined with --pmc, per
+    the pool rule.)"""
+    version, prof, probed = _find_profiler()
+    if prof is None:
+        pytest.skip("no rocprof binary found; probed: " + ", ".join(probed))
+    if version != "v3":
+        pytest.skip("memory-copy trace assertion needs rocprofv3")
+
+    nbytes = 77 * 1024 * 1024 + 4096  # distinctive size to find in the trace
+    out_dir = tmp_path / "prof"
+    out_dir.mkdir()
+    code = (
+        "import sys; sys.path.insert(0, %r); "
+        "import torch; torch.cuda.init(); "
+        "from covalent_ssh_plugin_amd.gpu import probe; "
+        "t = torch.ones(%d, dtype=torch.uint8, device='cuda'); "
+        "torch.cuda.synchronize(); "
+        "raw = probe.staged_d2h_bytes(t.data_ptr(), %d); "
+        "assert len(raw) == %d"
+    ) % (str(REPO), nbytes, nbytes, nbytes)
+    proc = subprocess.run(
+        [prof, "--kernel-trace", "--memory-copy-trace", "-d", str(out_dir),
+         "-o", "stage", "--", sys.executable, "-c", code],
+        cwd=str(out_dir),
+        env={**os.environ, "TMPDIR": "/tmp"},
+        capture_output=True,
+        text=True,
+        timeout=280,
+    )
+    assert proc.returncode == 0, (proc.stdout[-1000:], proc.stderr[-2000:])
+
+    import sqlite3
+
+    dbs = list(out_dir.rglob("*.db"))
+    assert dbs, "no rocprofv3 db produced"
+    found = False
+    for db in dbs:
+        con = sqlite3.connect(db)
+        tables = [
+            t for (t,) in con.execute(
+                "SELECT name FROM sqlite_master WHERE type='table'"
+            )
+            if "memory_copy" in t
+        ]
+        for table in tables:
+            cols = [c[1] for c in con.execute(f"PRAGMA table_info({table})")]
+            size_cols = [c for c in cols if c in ("size", "bytes", "size_bytes")]
+            if not size_cols:
+                continue
+            for (val,) in con.execute(f"SELECT {size_cols[0]} FROM {table}"):
+                if val == nbytes:
+                    found = True
+    assert found, f"no memory-copy record of {nbytes} bytes found in {dbs}"
+
+
+@pytest.mark.timeout(300)
 def test_probe_kernels_in_rocprof_trace(tmp_path):
     version, prof, probed = _find_profiler()
     if prof is None:
