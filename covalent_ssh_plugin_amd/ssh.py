@@ -545,16 +545,25 @@ class SSHExecutor(RemoteExecutor):
             return await transport.run(
                 self._submit_command(remote_script_file), env=env
             )
+        pidfile = self._pidfile(operation_id)
         cmd = (
             self._setsid_fragment(
-                self._submit_command(remote_script_file),
-                self._pidfile(operation_id),
+                self._submit_command(remote_script_file), pidfile
             )
             + "; exit $_csp_rc"
         )
-        self._inflight_fused[operation_id] = self._pidfile(operation_id)
+        self._inflight_fused[operation_id] = pidfile
         try:
-            return await transport.run(cmd, env=env)
+            return await transport.run(
+                cmd, env=env, timeout=self.task_timeout or None
+            )
+        except asyncio.TimeoutError:
+            # take the remote process group down with the timed-out wait
+            await self._kill_remote_group(transport, pidfile)
+            raise SSHTaskError(
+                f"task {operation_id} exceeded task_timeout="
+                f"{self.task_timeout}s"
+            )
         finally:
             self._inflight_fused.pop(operation_id, None)
 

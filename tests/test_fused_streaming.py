@@ -191,6 +191,32 @@ def test_cancel_unknown_task_raises(local_executor):
         asyncio.run(ex.cancel({"dispatch_id": "nope", "node_id": 0}))
 
 
+def test_template_task_timeout_kills_remote_group(local_executor):
+    """task_timeout on the DISCRETE template path (batch_roundtrips
+    off): the synchronous submit is bounded and the remote group dies."""
+    ex = local_executor(task_timeout=1.0, batch_roundtrips=False, poll_freq=1)
+    sleeper = _sleeper_factory()
+    pidfile = local_executor.home / ".cache" / "covalent" / "pid_tt_0"
+
+    async def go():
+        task = asyncio.create_task(
+            ex.execute(sleeper, [], {}, dispatch_id="tt", node_id=0)
+        )
+        for _ in range(200):
+            if pidfile.exists():
+                break
+            await asyncio.sleep(0.05)
+        pgid = int(pidfile.read_text().strip())
+        with pytest.raises(SSHTaskError, match="task_timeout"):
+            await task
+        return pgid
+
+    t0 = time.perf_counter()
+    pgid = asyncio.run(go())
+    assert time.perf_counter() - t0 < 12
+    assert _group_live_members(pgid) == []
+
+
 def test_fused_task_timeout_kills_remote_group(local_executor):
     """task_timeout on the fused path must also take down the remote
     process group (killing the local client alone leaves the task
