@@ -393,6 +393,73 @@ def test_isolated_gpu_electron(tmp_path):
     assert m1["staging"]["mode"] == "pinned", m1
 
 
+def test_cluster_two_nodes_over_sshim_gpu(tmp_path, sshim, monkeypatch):
+    """Multi-node fan-out on real hardware: two shim-SSH 'nodes' (two
+    fake remote homes on one box), a fan of bf16 matmul electrons spread
+    across both, plus transparent failover from a dead third node."""
+    monkeypatch.setenv(
+        "SSHIM_PASS_ENV",
+        "HSA_ENABLE_IPC_MODE_LEGACY,LD_LIBRARY_PATH,HIP_VISIBLE_DEVICES,"
+        "ROCR_VISIBLE_DEVICES,PYTORCH_ROCM_ARCH,TMPDIR",
+    )
+    from covalent_ssh_plugin_amd.cluster import SSHClusterExecutor
+
+    # the shim resolves SSHIM_HOME per invocation; with one home both
+    # "nodes" share a filesystem but keep separate transports, worker
+    # pools and slot tables (distinct hostnames -> distinct pool keys)
+    hosts = [
+        {"hostname": "deadnode.invalid"},  # always refuses: exercises failover
+        {"hostname": "gpu-node-a.test"},
+        {"hostname": "gpu-node-b.test"},
+    ]
+    cluster = SSHClusterExecutor(
+        hosts,
+        transport="ssh",
+        username="mi355x",
+        ssh_key_file=str(sshim.key),
+        cache_dir=str(tmp_path / "cache"),
+        python_path=sys.executable,
+        gpu_slots=max(1, torch.cuda.device_count()),
+        retry_connect=False,
+        pinned_staging_threshold_bytes=1024,
+    )
+
+    def electron(n):
+        import socket
+
+        import torch
+
+        a = torch.randn(n, n, device="cuda", dtype=torch.bfloat16)
+        b = torch.eye(n, device="cuda", dtype=torch.bfloat16)
+        c = a @ b
+        torch.cuda.synchronize()
+        return bool(torch.equal(c, a)), socket.gethostname()
+
+    async def go():
+        try:
+            return await asyncio.gather(
+                *[
+                    cluster.execute(electron, [512], {}, dispatch_id="cg", node_id=i)
+                    for i in range(6)
+                ]
+            )
+        finally:
+            from covalent_ssh_plugin_amd import SSHExecutor
+
+            await SSHExecutor.close_pool()
+
+    results = asyncio.run(go())
+    assert all(ok for ok, _ in results)
+    # the dead node was cooled down and every electron completed
+    assert cluster._unhealthy_until[0] > 0
+    stats = cluster.stats()
+    served = {h: s.get("count", 0) for h, s in stats.items()}
+    assert served.get("deadnode.invalid", 0) == 0
+    assert served.get("gpu-node-a.test", 0) + served.get("gpu-node-b.test", 0) == 6
+    assert served.get("gpu-node-a.test", 0) > 0
+    assert served.get("gpu-node-b.test", 0) > 0
+
+
 def test_probe_props_fast(gpu_lib):
     import time
 
