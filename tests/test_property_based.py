@@ -181,3 +181,41 @@ def test_channel_frame_roundtrip(tmp_path_factory, payloads):
             await ch.close()
 
     asyncio.run(main())
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    num_gpus=st.integers(min_value=1, max_value=4),
+    slots_per_gpu=st.integers(min_value=1, max_value=4),
+    ops=st.lists(st.integers(min_value=0, max_value=1), min_size=1, max_size=80),
+)
+def test_subslot_invariants(num_gpus, slots_per_gpu, ops):
+    """Oversubscription sub-slots: a (gpu, sub) pair is never held by
+    two electrons at once, sub ids stay in range, and worker tags are
+    unique across held slots (one worker process per sub-slot)."""
+
+    async def main():
+        table = SlotTable(num_gpus=num_gpus, slots_per_gpu=slots_per_gpu)
+        held = []
+        for op in ops:
+            if op == 0 and table.in_use < table.capacity:
+                held.append(await table.acquire())
+            elif held:
+                await held.pop().release()
+            pairs = [(s.gpu_id, s.sub_id) for s in held]
+            assert len(pairs) == len(set(pairs))
+            assert all(0 <= sub < slots_per_gpu for _, sub in pairs)
+            tags = [s.worker_tag for s in held]
+            assert len(tags) == len(set(tags))
+        for s in held:
+            await s.release()
+        assert table.in_use == 0
+        # full capacity restored with every sub-slot exactly once
+        final = [await table.acquire() for _ in range(table.capacity)]
+        assert sorted((s.gpu_id, s.sub_id) for s in final) == sorted(
+            (g, sub) for g in range(num_gpus) for sub in range(slots_per_gpu)
+        )
+        for s in final:
+            await s.release()
+
+    asyncio.run(main())
